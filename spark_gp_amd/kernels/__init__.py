@@ -1,0 +1,14 @@
+from .base import (ConstantTimesKernel, EyeKernel, Kernel, Scalar,
+                   ScalarTimesKernel, SumOfKernels,
+                   TrainableScalarTimesKernel,
+                   TrainingVectorsNotInitializedError, WhiteNoiseKernel,
+                   sqdist)
+from .compiled import CompiledKernel, compile_kernel
+from .rbf import ARDRBFKernel, RBFKernel
+
+__all__ = [
+    "Kernel", "EyeKernel", "WhiteNoiseKernel", "SumOfKernels",
+    "ScalarTimesKernel", "ConstantTimesKernel", "TrainableScalarTimesKernel",
+    "Scalar", "RBFKernel", "ARDRBFKernel", "sqdist",
+    "TrainingVectorsNotInitializedError", "CompiledKernel", "compile_kernel",
+]

@@ -1,0 +1,269 @@
+"""Batched PyTorch implementations of the GP compute kernels.
+
+This backend serves two roles:
+
+1. the *correctness oracle* — float64 on CPU, numerically equivalent to the
+   reference's per-expert Breeze math (``regression/GaussianProcessRegression.scala:55-68``,
+   ``classification/GaussianProcessClassifier.scala:74-129``);
+2. the fallback path on GPU for kernel trees the fused HIP kernels do not
+   cover.
+
+Everything is *batched over experts*: an ``[E, k, k]`` problem batch executes
+as a handful of tensor ops instead of E independent tasks (the reference runs
+one Spark task per expert).  Scalar accumulations (nll, gradient) are always
+done in float64 regardless of the compute dtype.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional, Tuple
+
+import numpy as np
+import torch
+
+from ..kernels.base import Kernel, sqdist
+from ..kernels.compiled import CompiledKernel
+
+
+def logdet_and_inv(K: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
+    """(logdet [...,], Kinv [..., k, k]) from one factorization per matrix.
+
+    Mirrors ``commons/util/logDetAndInv.scala`` (single LU there).  Here:
+    Cholesky (cheaper, PD-correct); experts whose Cholesky fails fall back to
+    LU-based slogdet+inv so non-PD iterates behave like the reference instead
+    of aborting (the reference's LU never fails on merely-indefinite K)."""
+    L, info = torch.linalg.cholesky_ex(K)
+    bad = info > 0
+    if bad.any():
+        logdet = torch.empty(K.shape[:-2], dtype=K.dtype, device=K.device)
+        Kinv = torch.empty_like(K)
+        good = ~bad
+        if good.any():
+            Lg = L[good]
+            logdet[good] = 2.0 * torch.log(Lg.diagonal(dim1=-2, dim2=-1)).sum(-1)
+            Kinv[good] = torch.cholesky_inverse(Lg)
+        Kb = K[bad]
+        _, logabsdet = torch.linalg.slogdet(Kb)
+        logdet[bad] = logabsdet
+        Kinv[bad] = torch.linalg.inv(Kb)
+        return logdet, Kinv
+    logdet = 2.0 * torch.log(L.diagonal(dim1=-2, dim2=-1)).sum(-1)
+    return logdet, torch.cholesky_inverse(L)
+
+
+# ---------------------------------------------------------------------------
+# Fused regression objective (compiled kernel canonical form)
+# ---------------------------------------------------------------------------
+
+def _base_matrices(cs: CompiledKernel, theta: np.ndarray, X: torch.Tensor):
+    """Kb = exp(-sq) for the canonical base; returns (Kb, sq_scaled, extras)."""
+    if cs.base == 'ard':
+        beta = torch.as_tensor(theta[cs.base_idx], dtype=X.dtype, device=X.device)
+        Xs = X * beta
+        sq = sqdist(Xs, Xs)
+    elif cs.base == 'rbf':
+        sigma = float(theta[cs.base_idx][0])
+        scale = 1.0 / (math.sqrt(2.0) * sigma)
+        Xs = X * scale
+        sq = sqdist(Xs, Xs)
+    else:
+        raise ValueError(f"unsupported base {cs.base}")
+    return torch.exp(-sq), sq, Xs
+
+
+def nll_grad_compiled(cs: CompiledKernel, theta: np.ndarray,
+                      X: torch.Tensor, y: torch.Tensor
+                      ) -> Tuple[float, np.ndarray]:
+    """Sum over the expert batch of the per-expert BCM negative log marginal
+    likelihood and its gradient w.r.t. the full hyperparameter vector.
+
+    nll_e = 1/2 y^T K^-1 y + 1/2 log|K|;  grad_i = -1/2 sum(dK_i * (aa^T - K^-1))
+    (``regression/GaussianProcessRegression.scala:55-68``), computed WITHOUT
+    materializing the [p, k, k] derivative tensor: the ARD/RBF derivative
+    contraction reduces to two GEMMs per batch (see K5 fusion plan,
+    SURVEY.md §2.4)."""
+    C = cs.amp(theta)
+    nu = cs.noise(theta)
+    Kb, sq, _ = _base_matrices(cs, theta, X)
+    k = X.shape[-2]
+    K = C * Kb + nu * torch.eye(k, dtype=X.dtype, device=X.device)
+
+    logdet, Kinv = logdet_and_inv(K)
+    alpha = (Kinv @ y.unsqueeze(-1)).squeeze(-1)                 # [E, k]
+    nll = (0.5 * (y * alpha).sum(-1).double() + 0.5 * logdet.double()).sum()
+
+    G = alpha.unsqueeze(-1) * alpha.unsqueeze(-2) - Kinv         # [E, k, k]
+    W0 = G * Kb                                                  # [E, k, k]
+
+    grad = np.zeros(cs.p)
+    if cs.amp_idx is not None:
+        grad[cs.amp_idx] = float(-0.5 * W0.double().sum())
+    if cs.base == 'ard':
+        beta = theta[cs.base_idx]
+        r = W0.sum(-1)                                           # [E, k]
+        t1 = (X * X * r.unsqueeze(-1)).sum(-2)                   # [E, d]
+        WX = W0 @ X                                              # [E, k, d]
+        t2 = (X * WX).sum(-2)                                    # [E, d]
+        gb = (C * (2.0 * t1 - 2.0 * t2).double().sum(0)).cpu().numpy() * beta
+        grad[cs.base_idx] = gb
+    else:  # rbf
+        sigma = float(theta[cs.base_idx][0])
+        # sum(W0 * sq_scaled) via the same two-GEMM contraction
+        Xs = X * (1.0 / (math.sqrt(2.0) * sigma))
+        r = W0.sum(-1)
+        t1 = (Xs * Xs * r.unsqueeze(-1)).sum(-2)
+        t2 = (Xs * (W0 @ Xs)).sum(-2)
+        s = float((2.0 * t1 - 2.0 * t2).double().sum())
+        grad[cs.base_idx.start] = -(C / sigma) * s
+    if cs.noise_idx:
+        trG = float(G.diagonal(dim1=-2, dim2=-1).double().sum())
+        for i in cs.noise_idx:
+            grad[i] += -0.5 * trG
+    return float(nll), grad
+
+
+# ---------------------------------------------------------------------------
+# Generic regression objective (any kernel tree; materialized derivatives)
+# ---------------------------------------------------------------------------
+
+def nll_grad_generic(kernel: Kernel, theta: np.ndarray,
+                     X: torch.Tensor, y: torch.Tensor
+                     ) -> Tuple[float, np.ndarray]:
+    kernel.set_hyperparameters(theta)
+    K, dK = kernel.training_kernel_and_derivative(X)   # [E,k,k], [E,p,k,k]
+    logdet, Kinv = logdet_and_inv(K)
+    alpha = (Kinv @ y.unsqueeze(-1)).squeeze(-1)
+    nll = (0.5 * (y * alpha).sum(-1).double() + 0.5 * logdet.double()).sum()
+    G = alpha.unsqueeze(-1) * alpha.unsqueeze(-2) - Kinv
+    grad_e = -0.5 * (dK * G.unsqueeze(-3)).sum((-1, -2))          # [E, p]
+    grad = grad_e.double().sum(0).cpu().numpy()
+    return float(nll), grad
+
+
+# ---------------------------------------------------------------------------
+# Batched Laplace approximation for GP classification
+# ---------------------------------------------------------------------------
+
+def laplace_nll_grad(kernel: Kernel, theta: np.ndarray,
+                     X: torch.Tensor, y: torch.Tensor, f: torch.Tensor,
+                     tol: float, max_newton_iter: int = 200
+                     ) -> Tuple[float, np.ndarray]:
+    """Batched Newton iteration (R&W Algorithm 3.1 with step halving) +
+    Algorithm 5.1 evidence/gradient, mirroring
+    ``classification/GaussianProcessClassifier.scala:74-129``.
+
+    ``f`` ([E, k]) is the per-expert latent vector, updated IN PLACE so it
+    warm-starts across L-BFGS evaluations (reference mutates cached RDD
+    state, ``GaussianProcessClassifier.scala:52-60``).
+
+    Experts converge at different Newton iteration counts: converged experts
+    are masked out of subsequent iterations (per-expert convergence mask —
+    SURVEY.md hard part #5)."""
+    kernel.set_hyperparameters(theta)
+    K, dK = kernel.training_kernel_and_derivative(X)   # [E,k,k], [E,p,k,k]
+    E, k = y.shape
+    dev, dt = X.device, X.dtype
+    eyek = torch.eye(k, dtype=dt, device=dev)
+
+    old_obj = torch.full((E,), -math.inf, dtype=torch.float64, device=dev)
+    new_obj = torch.full((E,), -torch.finfo(torch.float64).max,
+                         dtype=torch.float64, device=dev)
+    step = torch.ones(E, dtype=torch.float64, device=dev)
+    # final per-expert state needed by the evidence computation
+    L_out = torch.empty_like(K)
+    a_out = torch.empty_like(f)
+    pi_out = torch.empty_like(f)
+    sqw_out = torch.empty_like(f)
+
+    active = torch.ones(E, dtype=torch.bool, device=dev)
+    it = 0
+    while bool(active.any()) and it < max_newton_iter:
+        it += 1
+        idx = active.nonzero(as_tuple=True)[0]
+        Ki, fi, yi = K[idx], f[idx], y[idx]
+        pi = torch.sigmoid(fi)
+        w = pi * (1.0 - pi)
+        sqw = torch.sqrt(w)
+        B = eyek + sqw.unsqueeze(-1) * Ki * sqw.unsqueeze(-2)
+        Li = torch.linalg.cholesky(B)
+        grad_logp = yi - pi
+        b = w * fi + grad_logp
+        Kb = (Ki @ b.unsqueeze(-1)).squeeze(-1)
+        v = torch.cholesky_solve((sqw * Kb).unsqueeze(-1), Li).squeeze(-1)
+        a = b - sqw * v
+        si = step[idx].to(dt).unsqueeze(-1)
+        f_cand = (1.0 - si) * fi + si * (Ki @ a.unsqueeze(-1)).squeeze(-1)
+        obj_cand = (-0.5 * (a * f_cand).sum(-1).double()
+                    + torch.nn.functional.logsigmoid(
+                        (2.0 * yi - 1.0) * f_cand).double().sum(-1))
+
+        accept = obj_cand > old_obj[idx]
+        # store the state computed at the current f for every active expert —
+        # at loop exit it corresponds to the last iteration, as in the ref.
+        L_out[idx], a_out[idx] = Li, a
+        pi_out[idx], sqw_out[idx] = pi, sqw
+
+        acc_idx = idx[accept]
+        if acc_idx.numel():
+            f[acc_idx] = f_cand[accept]
+            old_obj[acc_idx] = new_obj[acc_idx]
+            new_obj[acc_idx] = obj_cand[accept]
+        rej_idx = idx[~accept]
+        if rej_idx.numel():
+            step[rej_idx] *= 0.5
+        still = (torch.abs(old_obj[idx] - new_obj[idx]) > tol) & (step[idx] > tol)
+        active[idx] = still
+
+    # ---- Algorithm 5.1: evidence and gradient --------------------------
+    L, a, pi, sqw = L_out, a_out, pi_out, sqw_out
+    logZ = new_obj - torch.log(L.diagonal(dim1=-2, dim2=-1)).double().sum(-1)
+
+    sqwD = torch.diag_embed(sqw)
+    R = sqw.unsqueeze(-1) * torch.cholesky_solve(sqwD, L)         # [E,k,k]
+    Cm = torch.linalg.solve_triangular(L, sqw.unsqueeze(-1) * K,
+                                       upper=False)               # [E,k,k]
+    d3 = -(2.0 * pi - 1.0) * pi * pi * torch.exp(-f)
+    diagK = K.diagonal(dim1=-2, dim2=-1)
+    diagCtC = (Cm * Cm).sum(-2)
+    s2 = -0.5 * (diagK - diagCtC) * d3                            # [E,k]
+
+    grad_logp = y - pi
+    p = dK.shape[-3]
+    grad = torch.zeros(p, dtype=torch.float64)
+    KR = K @ R                                                    # [E,k,k]
+    for i in range(p):
+        Di = dK[:, i]
+        s1 = (0.5 * (a.unsqueeze(-2) @ Di @ a.unsqueeze(-1)).squeeze(-1).squeeze(-1)
+              - 0.5 * (R * Di).sum((-1, -2)))
+        b = (Di @ grad_logp.unsqueeze(-1)).squeeze(-1)
+        s3 = b - (KR @ b.unsqueeze(-1)).squeeze(-1)
+        grad[i] = (s1.double() + (s2 * s3).sum(-1).double()).sum()
+
+    return float(-logZ.sum()), -grad.cpu().numpy()
+
+
+# ---------------------------------------------------------------------------
+# PPA accumulation (K12): K_mn K_nm and K_mn y over the local shard
+# ---------------------------------------------------------------------------
+
+def kmn_knm_and_kmny(kernel: Kernel, active: torch.Tensor,
+                     X: torch.Tensor, y: torch.Tensor,
+                     chunk_rows: int = 262144
+                     ) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Accumulate (K_mn K_nm [m,m], K_mn y [m]) over the local rows.
+
+    Mirrors ``commons/ProjectedGaussianProcessHelper.scala:20-36`` but
+    flattened: experts partition the rows, so the per-expert sum equals the
+    full-shard product.  Chunked so the [c, m] cross-kernel block never
+    exceeds a few hundred MB; cross-chunk accumulation is float64."""
+    m = active.shape[0]
+    n = X.shape[0]
+    KK = torch.zeros(m, m, dtype=torch.float64, device=X.device)
+    Ky = torch.zeros(m, dtype=torch.float64, device=X.device)
+    for s in range(0, n, chunk_rows):
+        e = min(n, s + chunk_rows)
+        A = kernel.cross_kernel(X[s:e], active)      # [c, m]
+        KK += (A.transpose(0, 1) @ A).double()
+        Ky += (A.transpose(0, 1) @ y[s:e].unsqueeze(-1)).squeeze(-1).double()
+    return KK, Ky
