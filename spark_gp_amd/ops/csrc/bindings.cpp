@@ -1,0 +1,131 @@
+// PyTorch bindings for the spark_gp_amd CDNA4 HIP kernels.
+
+#include <torch/extension.h>
+#include <hip/hip_runtime.h>
+#include <c10/hip/HIPStream.h>
+
+#include <vector>
+
+extern "C" hipError_t launch_fused_expert_nll(
+    const float* X, const float* y, const float* scale, float amp,
+    float noise, int E, int k, int d, double* out_nll, double* out_sumW0,
+    double* out_trG, double* out_contr, int* out_bad, hipStream_t stream,
+    size_t* lds_used);
+
+extern "C" hipError_t launch_cross_kernel_tile(
+    const float* X, const float* A, const float* s2v, float amp, int c,
+    int m, int d, void* out, int out_is_bf16, hipStream_t stream);
+
+extern "C" hipError_t launch_syrk_bf16(const void* Kc, int c, int m,
+                                       int split_k, float* KK,
+                                       hipStream_t stream);
+
+extern "C" hipError_t launch_colsum_gemv(const void* Kc, const float* y,
+                                         int c, int m, double* Ky,
+                                         hipStream_t stream);
+
+namespace {
+
+void check_hip(hipError_t err, const char* what) {
+  TORCH_CHECK(err == hipSuccess, what, ": ", hipGetErrorString(err));
+}
+
+hipStream_t current_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+}  // namespace
+
+// Returns (nll[E] f64, sumW0[E] f64, trG[E] f64, contr[E,d] f64, bad[E] i32)
+std::vector<torch::Tensor> fused_expert_nll(torch::Tensor X, torch::Tensor y,
+                                            torch::Tensor scale, double amp,
+                                            double noise) {
+  TORCH_CHECK(X.is_cuda() && X.dtype() == torch::kFloat32 && X.dim() == 3,
+              "X must be [E, k, d] float32 on GPU");
+  TORCH_CHECK(y.is_cuda() && y.dtype() == torch::kFloat32 && y.dim() == 2);
+  TORCH_CHECK(scale.is_cuda() && scale.dtype() == torch::kFloat32);
+  auto Xc = X.contiguous();
+  auto yc = y.contiguous();
+  auto sc = scale.contiguous();
+  const int E = X.size(0), k = X.size(1), d = X.size(2);
+  TORCH_CHECK(k <= 128 && d <= k, "fused_expert_nll requires k<=128, d<=k");
+  auto opts64 = torch::TensorOptions().dtype(torch::kFloat64).device(X.device());
+  auto opts32i = torch::TensorOptions().dtype(torch::kInt32).device(X.device());
+  auto nll = torch::empty({E}, opts64);
+  auto sumW0 = torch::empty({E}, opts64);
+  auto trG = torch::empty({E}, opts64);
+  auto contr = torch::empty({E, d}, opts64);
+  auto bad = torch::empty({E}, opts32i);
+  size_t lds = 0;
+  check_hip(launch_fused_expert_nll(
+                Xc.data_ptr<float>(), yc.data_ptr<float>(),
+                sc.data_ptr<float>(), (float)amp, (float)noise, E, k, d,
+                nll.data_ptr<double>(), sumW0.data_ptr<double>(),
+                trG.data_ptr<double>(), contr.data_ptr<double>(),
+                bad.data_ptr<int>(), current_stream(), &lds),
+            "fused_expert_nll");
+  return {nll, sumW0, trG, contr, bad};
+}
+
+bool fused_expert_nll_supported(int64_t k, int64_t d) {
+  if (k > 128 || d > k || k < 1) return false;
+  // LDS budget: mirror of nll_lds_bytes in expert_nll.hip
+  const int64_t kp = k + 1, dp = d + 1;
+  int64_t bytes = 8 * 256 + 4 * (2 * k * kp + k * dp + 4 * k + d) + 8 + 64;
+  return bytes <= 160 * 1024;
+}
+
+torch::Tensor cross_kernel_tile(torch::Tensor X, torch::Tensor A,
+                                torch::Tensor s2v, double amp, bool bf16_out) {
+  TORCH_CHECK(X.is_cuda() && X.dtype() == torch::kFloat32 && X.dim() == 2);
+  TORCH_CHECK(A.is_cuda() && A.dtype() == torch::kFloat32 && A.dim() == 2);
+  TORCH_CHECK(X.size(1) == A.size(1), "feature dims differ");
+  auto Xc = X.contiguous();
+  auto Ac = A.contiguous();
+  auto sc = s2v.contiguous();
+  const int c = X.size(0), m = A.size(0), d = X.size(1);
+  auto out = torch::empty(
+      {c, m}, torch::TensorOptions()
+                  .dtype(bf16_out ? torch::kBFloat16 : torch::kFloat32)
+                  .device(X.device()));
+  check_hip(launch_cross_kernel_tile(Xc.data_ptr<float>(), Ac.data_ptr<float>(),
+                                     sc.data_ptr<float>(), (float)amp, c, m, d,
+                                     out.data_ptr(), bf16_out ? 1 : 0,
+                                     current_stream()),
+            "cross_kernel_tile");
+  return out;
+}
+
+void syrk_bf16_acc(torch::Tensor Kc, torch::Tensor KK, int64_t split_k) {
+  TORCH_CHECK(Kc.is_cuda() && Kc.dtype() == torch::kBFloat16 && Kc.dim() == 2);
+  TORCH_CHECK(KK.is_cuda() && KK.dtype() == torch::kFloat32 && KK.dim() == 2);
+  auto Kcc = Kc.contiguous();
+  const int c = Kc.size(0), m = Kc.size(1);
+  TORCH_CHECK(KK.size(0) == m && KK.size(1) == m);
+  check_hip(launch_syrk_bf16(Kcc.data_ptr(), c, m, (int)split_k,
+                             KK.data_ptr<float>(), current_stream()),
+            "syrk_bf16");
+}
+
+void colsum_gemv_acc(torch::Tensor Kc, torch::Tensor y, torch::Tensor Ky) {
+  TORCH_CHECK(Kc.is_cuda() && Kc.dtype() == torch::kBFloat16 && Kc.dim() == 2);
+  TORCH_CHECK(y.is_cuda() && y.dtype() == torch::kFloat32);
+  TORCH_CHECK(Ky.is_cuda() && Ky.dtype() == torch::kFloat64);
+  auto Kcc = Kc.contiguous();
+  const int c = Kc.size(0), m = Kc.size(1);
+  check_hip(launch_colsum_gemv(Kcc.data_ptr(), y.contiguous().data_ptr<float>(),
+                               c, m, Ky.data_ptr<double>(), current_stream()),
+            "colsum_gemv");
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("fused_expert_nll", &fused_expert_nll,
+          "fused per-expert BCM nll+gradient primitives (CDNA4)");
+  mod.def("fused_expert_nll_supported", &fused_expert_nll_supported);
+  mod.def("cross_kernel_tile", &cross_kernel_tile,
+          "rectangular RBF/ARD kernel block (CDNA4)");
+  mod.def("syrk_bf16_acc", &syrk_bf16_acc,
+          "KK += Kc^T Kc, bf16 MFMA, fp32 accumulate (CDNA4)");
+  mod.def("colsum_gemv_acc", &colsum_gemv_acc,
+          "Ky += Kc^T y, fp64 accumulate (CDNA4)");
+}

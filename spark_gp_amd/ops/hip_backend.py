@@ -1,0 +1,134 @@
+"""Python wrapper over the hand-written CDNA4 HIP kernels (_hip_ext).
+
+Gradient assembly happens here (host-side chain rule over the kernel's raw
+contraction outputs), mirroring exactly the formulas of
+``torch_backend.nll_grad_compiled`` so the two backends are drop-in
+interchangeable and unit-diffable.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional, Tuple
+
+import numpy as np
+import torch
+
+from ..kernels.base import Kernel
+from ..kernels.compiled import CompiledKernel, compile_kernel
+from .. import _hip_ext as ext
+from . import torch_backend
+
+
+def _scale_vector(cs: CompiledKernel, theta: np.ndarray, d: int,
+                  device, dtype=torch.float32) -> torch.Tensor:
+    if cs.base == 'ard':
+        s = np.asarray(theta[cs.base_idx], dtype=np.float64)
+    else:  # rbf
+        sigma = float(theta[cs.base_idx][0])
+        s = np.full(d, 1.0 / (math.sqrt(2.0) * sigma))
+    return torch.as_tensor(s, dtype=dtype, device=device)
+
+
+def supports_nll(cs: CompiledKernel, X: torch.Tensor) -> bool:
+    if cs.base not in ("ard", "rbf"):
+        return False
+    if X.dtype != torch.float32 or X.dim() != 3:
+        return False
+    E, k, d = X.shape
+    return bool(ext.fused_expert_nll_supported(k, d))
+
+
+def nll_grad_compiled(cs: CompiledKernel, theta: np.ndarray,
+                      X: torch.Tensor, y: torch.Tensor
+                      ) -> Tuple[float, np.ndarray]:
+    E, k, d = X.shape
+    C = cs.amp(theta)
+    nu = cs.noise(theta)
+    scale = _scale_vector(cs, theta, d, X.device)
+    nll, sumW0, trG, contr, bad = ext.fused_expert_nll(
+        X, y.to(torch.float32), scale, float(C), float(nu))
+
+    bad_mask = bad != 0
+    n_bad = int(bad_mask.sum())
+    nll_total = float(nll.sum())
+    sumW0_t = float(sumW0.sum())
+    trG_t = float(trG.sum())
+    contr_t = contr.sum(0).cpu().numpy()              # [d]
+
+    grad = np.zeros(cs.p)
+    if cs.amp_idx is not None:
+        grad[cs.amp_idx] = -0.5 * sumW0_t
+    if cs.base == 'ard':
+        beta = np.asarray(theta[cs.base_idx], dtype=np.float64)
+        grad[cs.base_idx] = C * beta * contr_t
+    else:
+        sigma = float(theta[cs.base_idx][0])
+        grad[cs.base_idx.start] = -C / (2.0 * sigma ** 3) * float(contr_t.sum())
+    if cs.noise_idx:
+        for i in cs.noise_idx:
+            grad[i] += -0.5 * trG_t
+
+    if n_bad:
+        # fp32 Cholesky broke down for these experts (huge-amplitude
+        # iterates); recompute them on the torch path (LU fallback inside)
+        idx = bad_mask.nonzero(as_tuple=True)[0]
+        nll_b, grad_b = torch_backend.nll_grad_compiled(
+            cs, theta, X[idx], y[idx])
+        nll_total += nll_b
+        grad += grad_b
+    return nll_total, grad
+
+
+# ---------------------------------------------------------------------------
+# PPA path
+# ---------------------------------------------------------------------------
+
+def supports_ppa(kernel: Kernel, X: torch.Tensor) -> bool:
+    cs = compile_kernel(kernel)
+    return (cs is not None and cs.base in ("ard", "rbf")
+            and X.dtype == torch.float32)
+
+
+def _s2_vector(cs: CompiledKernel, theta: np.ndarray, d: int, device):
+    s = _scale_vector(cs, theta, d, device, dtype=torch.float64)
+    return (s * s).to(torch.float32)
+
+
+def kmn_knm_and_kmny(kernel: Kernel, active: torch.Tensor,
+                     X: torch.Tensor, y: torch.Tensor,
+                     chunk_rows: int = 131072
+                     ) -> Tuple[torch.Tensor, torch.Tensor]:
+    cs = compile_kernel(kernel)
+    theta = kernel.get_hyperparameters()
+    n, d = X.shape
+    m = active.shape[0]
+    C = cs.amp(theta)
+    s2 = _s2_vector(cs, theta, d, X.device)
+    act32 = active.to(torch.float32).contiguous()
+    y32 = y.to(torch.float32)
+
+    KK = torch.zeros(m, m, dtype=torch.float32, device=X.device)
+    Ky = torch.zeros(m, dtype=torch.float64, device=X.device)
+    ntile = (m + 127) // 128
+    tiles = ntile * (ntile + 1) // 2
+    split_k = max(1, min(64, (512 + tiles - 1) // tiles))
+    for s in range(0, n, chunk_rows):
+        e = min(n, s + chunk_rows)
+        Kc = ext.cross_kernel_tile(X[s:e].contiguous(), act32, s2, float(C),
+                                   True)
+        ext.syrk_bf16_acc(Kc, KK, split_k)
+        ext.colsum_gemv_acc(Kc, y32[s:e].contiguous(), Ky)
+    return KK.double(), Ky
+
+
+def cross_kernel(kernel: Kernel, Xtest: torch.Tensor,
+                 Xtrain: torch.Tensor) -> torch.Tensor:
+    cs = compile_kernel(kernel)
+    theta = kernel.get_hyperparameters()
+    d = Xtest.shape[-1]
+    C = cs.amp(theta)
+    s2 = _s2_vector(cs, theta, d, Xtest.device)
+    return ext.cross_kernel_tile(Xtest.to(torch.float32).contiguous(),
+                                 Xtrain.to(torch.float32).contiguous(),
+                                 s2, float(C), False).to(Xtest.dtype)
