@@ -42,6 +42,11 @@ def _allow_fallback() -> bool:
     return os.environ.get("SPARK_GP_AMD_ALLOW_TORCH_FALLBACK", "0") == "1"
 
 
+def _force_torch() -> bool:
+    """Debug/A-B switch: run the eager torch path even when HIP is present."""
+    return os.environ.get("SPARK_GP_AMD_FORCE_TORCH", "0") == "1"
+
+
 def hip_available() -> bool:
     return _load_hip() is not None
 
@@ -63,7 +68,7 @@ def _require_hip_or_fallback(what: str) -> bool:
 def nll_grad_compiled(cs: CompiledKernel, theta: np.ndarray,
                       X: torch.Tensor, y: torch.Tensor
                       ) -> Tuple[float, np.ndarray]:
-    if X.is_cuda and cs.base in ("ard", "rbf"):
+    if X.is_cuda and cs.base in ("ard", "rbf") and not _force_torch():
         hip = _load_hip()
         if _require_hip_or_fallback("nll_grad_compiled") and \
                 hip.supports_nll(cs, X):
@@ -86,7 +91,7 @@ def laplace_nll_grad(kernel: Kernel, theta: np.ndarray, X: torch.Tensor,
 def kmn_knm_and_kmny(kernel: Kernel, active: torch.Tensor,
                      X: torch.Tensor, y: torch.Tensor,
                      chunk_rows: int = 262144):
-    if X.is_cuda:
+    if X.is_cuda and not _force_torch():
         hip = _load_hip()
         if _require_hip_or_fallback("kmn_knm_and_kmny") and \
                 hip.supports_ppa(kernel, X):

@@ -14,10 +14,11 @@ extern "C" hipError_t launch_fused_expert_nll(
 
 extern "C" hipError_t launch_cross_kernel_tile(
     const float* X, const float* A, const float* s2v, float amp, int c,
-    int m, int d, void* out, int out_is_bf16, hipStream_t stream);
+    int m, int d, void* out, void* out_lo, int out_is_bf16,
+    hipStream_t stream);
 
-extern "C" hipError_t launch_syrk_bf16(const void* Kc, int c, int m,
-                                       int split_k, float* KK,
+extern "C" hipError_t launch_syrk_bf16(const void* Kc, const void* Kl,
+                                       int c, int m, int split_k, float* KK,
                                        hipStream_t stream);
 
 extern "C" hipError_t launch_colsum_gemv(const void* Kc, const float* y,
@@ -75,35 +76,49 @@ bool fused_expert_nll_supported(int64_t k, int64_t d) {
   return bytes <= 160 * 1024;
 }
 
-torch::Tensor cross_kernel_tile(torch::Tensor X, torch::Tensor A,
-                                torch::Tensor s2v, double amp, bool bf16_out) {
+std::vector<torch::Tensor> cross_kernel_tile(torch::Tensor X, torch::Tensor A,
+                                             torch::Tensor s2v, double amp,
+                                             bool bf16_out, bool hilo) {
   TORCH_CHECK(X.is_cuda() && X.dtype() == torch::kFloat32 && X.dim() == 2);
   TORCH_CHECK(A.is_cuda() && A.dtype() == torch::kFloat32 && A.dim() == 2);
   TORCH_CHECK(X.size(1) == A.size(1), "feature dims differ");
+  TORCH_CHECK(!hilo || bf16_out, "hilo requires bf16 output");
   auto Xc = X.contiguous();
   auto Ac = A.contiguous();
   auto sc = s2v.contiguous();
   const int c = X.size(0), m = A.size(0), d = X.size(1);
-  auto out = torch::empty(
-      {c, m}, torch::TensorOptions()
+  auto opts = torch::TensorOptions()
                   .dtype(bf16_out ? torch::kBFloat16 : torch::kFloat32)
-                  .device(X.device()));
+                  .device(X.device());
+  auto out = torch::empty({c, m}, opts);
+  torch::Tensor lo;
+  if (hilo) lo = torch::empty({c, m}, opts);
   check_hip(launch_cross_kernel_tile(Xc.data_ptr<float>(), Ac.data_ptr<float>(),
                                      sc.data_ptr<float>(), (float)amp, c, m, d,
-                                     out.data_ptr(), bf16_out ? 1 : 0,
-                                     current_stream()),
+                                     out.data_ptr(),
+                                     hilo ? lo.data_ptr() : nullptr,
+                                     bf16_out ? 1 : 0, current_stream()),
             "cross_kernel_tile");
-  return out;
+  if (hilo) return {out, lo};
+  return {out};
 }
 
-void syrk_bf16_acc(torch::Tensor Kc, torch::Tensor KK, int64_t split_k) {
+void syrk_bf16_acc(torch::Tensor Kc, c10::optional<torch::Tensor> Kl,
+                   torch::Tensor KK, int64_t split_k) {
   TORCH_CHECK(Kc.is_cuda() && Kc.dtype() == torch::kBFloat16 && Kc.dim() == 2);
   TORCH_CHECK(KK.is_cuda() && KK.dtype() == torch::kFloat32 && KK.dim() == 2);
   auto Kcc = Kc.contiguous();
   const int c = Kc.size(0), m = Kc.size(1);
   TORCH_CHECK(KK.size(0) == m && KK.size(1) == m);
-  check_hip(launch_syrk_bf16(Kcc.data_ptr(), c, m, (int)split_k,
-                             KK.data_ptr<float>(), current_stream()),
+  torch::Tensor Klc;
+  if (Kl.has_value()) {
+    TORCH_CHECK(Kl->sizes() == Kc.sizes() && Kl->dtype() == torch::kBFloat16);
+    Klc = Kl->contiguous();
+  }
+  check_hip(launch_syrk_bf16(Kcc.data_ptr(),
+                             Kl.has_value() ? Klc.data_ptr() : nullptr, c, m,
+                             (int)split_k, KK.data_ptr<float>(),
+                             current_stream()),
             "syrk_bf16");
 }
 

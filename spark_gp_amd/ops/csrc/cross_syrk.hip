@@ -35,6 +35,8 @@ cross_kernel_tile_kernel(const float* __restrict__ X,    // [c, d]
                          const float amp,
                          const int c, const int m, const int d,
                          bf16* __restrict__ out_bf,      // [c, m] or null
+                         bf16* __restrict__ out_lo,      // [c, m] or null:
+                                                         // residual v - hi
                          float* __restrict__ out_f32) {  // [c, m] or null
   __shared__ float xs[CK_TILE][CK_DBLK + 1];
   __shared__ float as[CK_TILE][CK_DBLK + 1];
@@ -93,8 +95,15 @@ cross_kernel_tile_kernel(const float* __restrict__ X,    // [c, d]
       const int gc = col0 + tc + j;
       if (gc >= m) continue;
       const float v = amp * __expf(-acc[i][j]);
-      if (out_bf) out_bf[(size_t)gr * m + gc] = (bf16)v;
-      else out_f32[(size_t)gr * m + gc] = v;
+      if (out_bf) {
+        const bf16 hi = (bf16)v;
+        out_bf[(size_t)gr * m + gc] = hi;
+        // two-term bf16 split: hi + lo carries ~16 mantissa bits, so the
+        // SYRK's input-quantization error drops to fp32 class
+        if (out_lo) out_lo[(size_t)gr * m + gc] = (bf16)(v - (float)hi);
+      } else {
+        out_f32[(size_t)gr * m + gc] = v;
+      }
     }
   }
 }
@@ -112,13 +121,18 @@ cross_kernel_tile_kernel(const float* __restrict__ X,    // [c, d]
 #define SY_BK 32
 #define SY_STR 40     // LDS row stride in bf16 elements (80 B)
 
-extern "C" __global__ void __launch_bounds__(256)
-syrk_bf16_kernel(const bf16* __restrict__ Kc,   // [c, m]
+template <bool HILO>
+__global__ void __launch_bounds__(256)
+syrk_bf16_kernel(const bf16* __restrict__ Kc,   // [c, m] hi part
+                 const bf16* __restrict__ Kl,   // [c, m] lo part (HILO only)
                  const int c, const int m, const int ntile,
                  const int split_k,
                  float* __restrict__ KK) {      // [m, m] accumulated
+  // with HILO: KK += hi^T hi + hi^T lo + lo^T hi  (lo^T lo ~ 2^-32, dropped)
   __shared__ __align__(16) bf16 lt[128 * SY_STR];
   __shared__ __align__(16) bf16 rt[128 * SY_STR];
+  __shared__ __align__(16) bf16 ltl[HILO ? 128 * SY_STR : 1];
+  __shared__ __align__(16) bf16 rtl[HILO ? 128 * SY_STR : 1];
 
   const int tile = blockIdx.x;
   const int ti = tile / ntile, tj = tile % ntile;
@@ -158,32 +172,27 @@ syrk_bf16_kernel(const bf16* __restrict__ Kc,   // [c, m]
       const int col = seg * 8;
       const int gk = krow0 + kk;
       bf16 vals[8];
-      if (gk < c) {
-        const bf16* src = Kc + (size_t)gk * m;
+      auto stage = [&](const bf16* mat, int c0, bf16* dst) {
+        if (gk < c) {
+          const bf16* src = mat + (size_t)gk * m;
 #pragma unroll
-        for (int u = 0; u < 8; ++u) {
-          const int gc = i0 + col + u;
-          vals[u] = (gc < m) ? src[gc] : (bf16)0.f;
+          for (int u = 0; u < 8; ++u) {
+            const int gc = c0 + col + u;
+            vals[u] = (gc < m) ? src[gc] : (bf16)0.f;
+          }
+        } else {
+#pragma unroll
+          for (int u = 0; u < 8; ++u) vals[u] = (bf16)0.f;
         }
-      } else {
 #pragma unroll
-        for (int u = 0; u < 8; ++u) vals[u] = (bf16)0.f;
+        for (int u = 0; u < 8; ++u) dst[(col + u) * SY_STR + kk] = vals[u];
+      };
+      stage(Kc, i0, lt);
+      stage(Kc, j0, rt);
+      if (HILO) {
+        stage(Kl, i0, ltl);
+        stage(Kl, j0, rtl);
       }
-#pragma unroll
-      for (int u = 0; u < 8; ++u) lt[(col + u) * SY_STR + kk] = vals[u];
-      if (gk < c) {
-        const bf16* src = Kc + (size_t)gk * m;
-#pragma unroll
-        for (int u = 0; u < 8; ++u) {
-          const int gc = j0 + col + u;
-          vals[u] = (gc < m) ? src[gc] : (bf16)0.f;
-        }
-      } else {
-#pragma unroll
-        for (int u = 0; u < 8; ++u) vals[u] = (bf16)0.f;
-      }
-#pragma unroll
-      for (int u = 0; u < 8; ++u) rt[(col + u) * SY_STR + kk] = vals[u];
     }
     __syncthreads();
 
@@ -194,12 +203,21 @@ syrk_bf16_kernel(const bf16* __restrict__ Kc,   // [c, m]
       for (int a = 0; a < 4; ++a) {
         const int arow = wr + a * 16 + l16;
         bf16x8 afrag = *(const bf16x8*)&lt[arow * SY_STR + kgrp * 8];
+        bf16x8 afl;
+        if (HILO) afl = *(const bf16x8*)&ltl[arow * SY_STR + kgrp * 8];
 #pragma unroll
         for (int b = 0; b < 4; ++b) {
           const int bcol = wc + b * 16 + l16;
           bf16x8 bfrag = *(const bf16x8*)&rt[bcol * SY_STR + kgrp * 8];
           acc[a][b] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afrag, bfrag, acc[a][b], 0, 0, 0);
+          if (HILO) {
+            bf16x8 bfl = *(const bf16x8*)&rtl[bcol * SY_STR + kgrp * 8];
+            acc[a][b] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afrag, bfl, acc[a][b], 0, 0, 0);
+            acc[a][b] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afl, bfrag, acc[a][b], 0, 0, 0);
+          }
         }
       }
     }
@@ -260,22 +278,30 @@ colsum_gemv_kernel(const bf16* __restrict__ Kc,   // [c, m]
 // ---------------------------------------------------------------------------
 extern "C" hipError_t launch_cross_kernel_tile(
     const float* X, const float* A, const float* s2v, float amp,
-    int c, int m, int d, void* out, int out_is_bf16, hipStream_t stream) {
+    int c, int m, int d, void* out, void* out_lo, int out_is_bf16,
+    hipStream_t stream) {
   dim3 grid((c + CK_TILE - 1) / CK_TILE, (m + CK_TILE - 1) / CK_TILE);
   hipLaunchKernelGGL(cross_kernel_tile_kernel, grid, dim3(256), 0, stream,
                      X, A, s2v, amp, c, m, d,
                      out_is_bf16 ? (bf16*)out : nullptr,
+                     out_is_bf16 ? (bf16*)out_lo : nullptr,
                      out_is_bf16 ? nullptr : (float*)out);
   return hipGetLastError();
 }
 
-extern "C" hipError_t launch_syrk_bf16(const void* Kc, int c, int m,
-                                       int split_k, float* KK,
+extern "C" hipError_t launch_syrk_bf16(const void* Kc, const void* Kl,
+                                       int c, int m, int split_k, float* KK,
                                        hipStream_t stream) {
   const int ntile = (m + 127) / 128;
   dim3 grid(ntile * ntile, split_k);
-  hipLaunchKernelGGL(syrk_bf16_kernel, grid, dim3(256), 0, stream,
-                     (const bf16*)Kc, c, m, ntile, split_k, KK);
+  if (Kl) {
+    hipLaunchKernelGGL((syrk_bf16_kernel<true>), grid, dim3(256), 0, stream,
+                       (const bf16*)Kc, (const bf16*)Kl, c, m, ntile,
+                       split_k, KK);
+  } else {
+    hipLaunchKernelGGL((syrk_bf16_kernel<false>), grid, dim3(256), 0, stream,
+                       (const bf16*)Kc, nullptr, c, m, ntile, split_k, KK);
+  }
   return hipGetLastError();
 }
 

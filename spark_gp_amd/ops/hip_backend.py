@@ -115,10 +115,15 @@ def kmn_knm_and_kmny(kernel: Kernel, active: torch.Tensor,
     split_k = max(1, min(64, (512 + tiles - 1) // tiles))
     for s in range(0, n, chunk_rows):
         e = min(n, s + chunk_rows)
-        Kc = ext.cross_kernel_tile(X[s:e].contiguous(), act32, s2, float(C),
-                                   True)
-        ext.syrk_bf16_acc(Kc, KK, split_k)
-        ext.colsum_gemv_acc(Kc, y32[s:e].contiguous(), Ky)
+        # hi/lo bf16 split: KK += hi^T hi + hi^T lo + lo^T hi keeps the
+        # input-quantization error at fp32 level so the PD matrix's small
+        # eigenvalues (~sigma2 * lambda_min(Kmm)) survive
+        Kc, Kl = ext.cross_kernel_tile(X[s:e].contiguous(), act32, s2,
+                                       float(C), True, True)
+        ext.syrk_bf16_acc(Kc, Kl, KK, split_k)
+        yc = y32[s:e].contiguous()
+        ext.colsum_gemv_acc(Kc, yc, Ky)
+        ext.colsum_gemv_acc(Kl, yc, Ky)
     return KK.double(), Ky
 
 
@@ -131,4 +136,4 @@ def cross_kernel(kernel: Kernel, Xtest: torch.Tensor,
     s2 = _s2_vector(cs, theta, d, Xtest.device)
     return ext.cross_kernel_tile(Xtest.to(torch.float32).contiguous(),
                                  Xtrain.to(torch.float32).contiguous(),
-                                 s2, float(C), False).to(Xtest.dtype)
+                                 s2, float(C), False, False)[0].to(Xtest.dtype)

@@ -54,14 +54,33 @@ def magic_vector_matrix(kernel: Kernel, KK: torch.Tensor, Ky: torch.Tensor,
     nu = kernel.white_noise_var()
     PD = nu * Kmm + KK
 
-    Lpd, info = torch.linalg.cholesky_ex(PD)
-    if int(info) > 0:
-        raise NotPositiveDefiniteError()
+    Lpd = _chol_with_jitter(PD)
     magic_vector = torch.cholesky_solve(Ky.unsqueeze(-1), Lpd).squeeze(-1)
-
-    Lmm, info_mm = torch.linalg.cholesky_ex(Kmm)
-    if int(info_mm) > 0:
-        raise NotPositiveDefiniteError()
+    Lmm = _chol_with_jitter(Kmm)
     magic_matrix = (torch.cholesky_inverse(Lpd) * nu
                     - torch.cholesky_inverse(Lmm))
     return magic_vector, magic_matrix
+
+
+def _chol_with_jitter(M: torch.Tensor, max_tries: int = 6) -> torch.Tensor:
+    """Cholesky with an escalating-jitter ladder.
+
+    The reference asserts PD via eigSym and throws immediately
+    (``ProjectedGaussianProcessHelper.scala:62-65``).  Here, matrices that are
+    PD in exact arithmetic can be numerically indefinite (fp32/bf16
+    accumulation of K_mn K_nm on the GPU path), so failures retry with
+    jitter = eps_rel * mean(diag) escalating 1e-12 .. 1e-2 before raising —
+    a strict robustness superset of the reference behavior."""
+    L, info = torch.linalg.cholesky_ex(M)
+    if int(info) == 0:
+        return L
+    scale = float(M.diagonal().abs().mean())
+    eps = 1e-12
+    for _ in range(max_tries):
+        jit = eps * scale * torch.eye(M.shape[-1], dtype=M.dtype,
+                                      device=M.device)
+        L, info = torch.linalg.cholesky_ex(M + jit)
+        if int(info) == 0:
+            return L
+        eps *= 100.0
+    raise NotPositiveDefiniteError()
