@@ -42,6 +42,9 @@ def parse_args():
     p.add_argument("--sigma2", type=float, default=1e-3)
     p.add_argument("--seed", type=int, default=13)
     p.add_argument("--device", type=str, default=None)
+    p.add_argument("--ppa-precision", type=str, default="mixed",
+                   choices=["mixed", "fp64"],
+                   help="PPA SYRK path: hi/lo bf16 MFMA (mixed) or fp64")
     return p.parse_args()
 
 
@@ -75,6 +78,7 @@ def main():
                 .setSigma2(args.sigma2)
                 .setMaxIter(args.max_iter)
                 .setSeed(args.seed)
+                .setPpaPrecision(args.ppa_precision)
                 .setDevice(str(device)))
 
     def sync():
@@ -119,6 +123,7 @@ def main():
                 "dim": args.dim,
                 "active_set": args.active_set,
                 "expert_size": args.expert_size,
+                "ppa_precision": args.ppa_precision,
                 "parallelism": f"expert-parallel dp{world}",
             },
         }

@@ -77,6 +77,8 @@ class GaussianProcessParams:
         self._seed = 0
         self._device: Optional[str] = None        # None -> auto
         self._dtype: Optional[torch.dtype] = None  # None -> f64 CPU / f32 GPU
+        self._ppa_precision = "fp64"  # 'fp64' (reference-parity) | 'mixed'
+                                      # (hi/lo bf16 MFMA SYRK, fastest)
 
     # Reference-parity camelCase setters -------------------------------
     def setKernel(self, factory: Callable[[], Kernel]):
@@ -114,6 +116,15 @@ class GaussianProcessParams:
     # Additive (no reference analog)
     def setDevice(self, device: str):
         self._device = device
+        return self
+
+    def setPpaPrecision(self, p: str):
+        """'fp64' (default; matches the reference's double-precision PPA) or
+        'mixed' (hi/lo-split bf16 MFMA SYRK with fp32 accumulation — the
+        fastest path; fine when sigma2 is not tiny)."""
+        if p not in ("fp64", "mixed"):
+            raise ValueError("ppa precision must be 'fp64' or 'mixed'")
+        self._ppa_precision = p
         return self
 
     def setDtype(self, dtype: torch.dtype):
@@ -198,7 +209,8 @@ class GaussianProcessCommons(GaussianProcessParams):
         instr.log_timing("active_set", time.perf_counter() - t0)
 
         t0 = time.perf_counter()
-        KK, Ky = accumulate_ppa_stats(kernel, active, X, y, comm)
+        KK, Ky = accumulate_ppa_stats(kernel, active, X, y, comm,
+                                      precision=self._ppa_precision)
         instr.log_timing("ppa_accumulate", time.perf_counter() - t0)
 
         t0 = time.perf_counter()

@@ -90,12 +90,13 @@ def laplace_nll_grad(kernel: Kernel, theta: np.ndarray, X: torch.Tensor,
 
 def kmn_knm_and_kmny(kernel: Kernel, active: torch.Tensor,
                      X: torch.Tensor, y: torch.Tensor,
-                     chunk_rows: int = 262144):
+                     chunk_rows: int = 262144, precision: str = "fp64"):
     if X.is_cuda and not _force_torch():
         hip = _load_hip()
         if _require_hip_or_fallback("kmn_knm_and_kmny") and \
                 hip.supports_ppa(kernel, X):
-            return hip.kmn_knm_and_kmny(kernel, active, X, y, chunk_rows)
+            return hip.kmn_knm_and_kmny(kernel, active, X, y, chunk_rows,
+                                        precision)
     return torch_backend.kmn_knm_and_kmny(kernel, active, X, y, chunk_rows)
 
 

@@ -69,10 +69,11 @@ std::vector<torch::Tensor> fused_expert_nll(torch::Tensor X, torch::Tensor y,
 }
 
 bool fused_expert_nll_supported(int64_t k, int64_t d) {
-  if (k > 128 || d > k || k < 1) return false;
-  // LDS budget: mirror of nll_lds_bytes in expert_nll.hip
-  const int64_t kp = k + 1, dp = d + 1;
-  int64_t bytes = 8 * 256 + 4 * (2 * k * kp + k * dp + 4 * k + d) + 8 + 64;
+  if (k > 128 || d > k || d > 64 || k < 1) return false;
+  // LDS budget: mirror of nll_lds_bytes2 in expert_nll.hip
+  const int64_t tsz = std::max<int64_t>(k * 33, 32 * (k + 1));
+  int64_t bytes = 8 * 10 + 4 * (k * (k + 1) + tsz + k * (d + 1) + 4 * k + d)
+                  + 16;
   return bytes <= 160 * 1024;
 }
 

@@ -1,105 +1,119 @@
 // Fused per-expert BCM negative-log-marginal-likelihood + gradient kernel
-// for MI355X (gfx950, CDNA4).
+// for MI355X (gfx950, CDNA4) — v2: blocked, in-place, single-LDS-buffer.
 //
-// One workgroup per expert; the whole per-expert pipeline that the reference
-// runs as a chain of Breeze/LAPACK calls per Spark task
-// (regression/GaussianProcessRegression.scala:55-68 — kernel build, LU,
-// inverse, alpha, trace products; kernel/ARDRBFKernel.scala:48-79 — K and
-// d derivative matrices) executes in LDS without touching HBM between
-// stages, and WITHOUT ever materializing the [d, k, k] derivative tensor:
+// One workgroup (512 threads = 8 waves) per expert; the whole per-expert
+// pipeline that the reference runs as a chain of Breeze/LAPACK calls per
+// Spark task (regression/GaussianProcessRegression.scala:55-68, kernel
+// build + LU + inverse + alpha + trace products; kernel/ARDRBFKernel.scala:
+// 48-79 — K and d derivative matrices) executes in LDS without touching HBM
+// between stages, and WITHOUT materializing the [d, k, k] derivative tensor:
 //
-//   1. stage X[e] (k x d) in LDS
-//   2. K = amp * exp(-sum_d s2_d (x_ad - x_bd)^2) + noise * I     (LDS)
-//   3. Cholesky K = L L^T in place (fp32, fp64 logdet)
-//   4. V = L^-1 (column-parallel trtri into a second LDS buffer)
-//   5. alpha = V^T (V y);   nll = 1/2 y.alpha + 1/2 logdet
-//   6. W0 = (alpha alpha^T - V^T V) o Kb   (Kb recomputed elementwise; trG
-//      and sum(W0) accumulated on the fly)
-//   7. gradient contraction per input dim j (the K5 fusion):
-//        contr_j = sum_ab W0_ab (x_aj - x_bj)^2
-//                = 2 sum_a x_aj^2 r_a - 2 sum_a x_aj (W0 X)_aj
-//      i.e. one k x d GEMM + elementwise work instead of d full k x k
-//      derivative matrices.
+//   B  K = amp * exp(-sum_d s2_d (x_ad - x_bd)^2) + noise I   (lower, LDS)
+//   C  blocked right-looking Cholesky, IN PLACE: per 32-column block, the
+//      diagonal block is factored AND inverted by one wave (lockstep
+//      __syncwarp steps, no workgroup barriers), the panel solve becomes a
+//      dense GEMM against the inverted diagonal, the trailing update is an
+//      all-thread data-parallel GEMM.  fp32; fp64 logdet on the fly.
+//   D  blocked in-place triangular inverse of the off-diagonal blocks
+//      (right-to-left column blocks; every row is independent, fully
+//      parallel — the identity V_IJ = -(sum_{K>J} V_IK L_KJ) L_JJ^-1).
+//   E  alpha = V^T (V y);  nll = 1/2 y.alpha + 1/2 logdet
+//   L  in-place lauum: K^-1 = V^T V (ascending row blocks through a temp
+//      row buffer)
+//   W  W0 = (alpha alpha^T - K^-1) o Kb, Kb recomputed elementwise from X;
+//      trG / sum(W0) accumulated on the fly
+//   H  gradient contraction per input dim (the K5 fusion):
+//      contr_j = 2 sum_a x_aj^2 r_a - 2 sum_a x_aj (W0 X)_aj
 //
 // Host-side chain rule turns (contr, sumW0, trG) into the gradient for both
-// the ARD-RBF and RBF parameterizations (ops/hip_backend.py).
+// ARD-RBF and RBF (ops/hip_backend.py).  Single k x (k+1) working buffer +
+// one k x 33 temp => ~69 KB LDS at k=100, two experts resident per CU.
 //
 // Numerics: fp32 storage/factorization, fp64 scalar accumulation.  Experts
 // whose fp32 Cholesky breaks down are flagged in out_bad and recomputed on
-// the torch fallback path by the host (mirrors the reference's LU tolerance
-// of indefinite iterates, commons/util/logDetAndInv.scala).
+// the torch fallback path by the host.
 //
-// Constraints: k <= 128, d <= k, LDS budget checked host-side.
+// Constraints: k <= 128, d <= min(k, 64); LDS budget checked host-side.
 
 #include <hip/hip_runtime.h>
 #include <math.h>
 
-#define WG 256
+#define WG 512
+#define NB 32
 
-// dynamic LDS layout helper
 struct NllLds {
-  float* A;     // k * kp   (K -> L -> W0)
-  float* V;     // k * kp   (L^-1, later W0*X)
-  float* X;     // k * dp   raw features
+  float* A;     // k * (k+1)  K -> L/V -> K^-1 -> W0
+  float* T;     // temp: max(k*33, 32*(k+1))
+  float* X;     // k * (d+1) raw features
   float* yb;    // k
   float* alpha; // k
   float* tvec;  // k
   float* rrow;  // k
   float* s2;    // d
-  double* red;  // WG (reduction scratch)
+  double* red;  // 8 (per-wave partials)
+  double* misc; // 2: logdet, scratch
   int* bad;     // 1
 };
 
-__device__ inline NllLds carve_lds(char* base, int k, int d, int kp, int dp) {
-  NllLds L;
+static __host__ __device__ inline size_t nll_lds_bytes2(int k, int d) {
+  size_t tsz = (size_t)(k * 33 > 32 * (k + 1) ? k * 33 : 32 * (k + 1));
   size_t off = 0;
-  auto take = [&](size_t bytes, size_t align) {
-    off = (off + align - 1) & ~(align - 1);
-    size_t o = off; off += bytes; return o;
-  };
-  L.red   = (double*)(base + take(sizeof(double) * WG, 8));
-  L.A     = (float*)(base + take(sizeof(float) * k * kp, 4));
-  L.V     = (float*)(base + take(sizeof(float) * k * kp, 4));
-  L.X     = (float*)(base + take(sizeof(float) * k * dp, 4));
-  L.yb    = (float*)(base + take(sizeof(float) * k, 4));
-  L.alpha = (float*)(base + take(sizeof(float) * k, 4));
-  L.tvec  = (float*)(base + take(sizeof(float) * k, 4));
-  L.rrow  = (float*)(base + take(sizeof(float) * k, 4));
-  L.s2    = (float*)(base + take(sizeof(float) * d, 4));
-  L.bad   = (int*)(base + take(sizeof(int), 4));
-  return L;
-}
-
-__host__ size_t nll_lds_bytes(int k, int d, int kp, int dp) {
-  size_t off = 0;
-  auto take = [&](size_t bytes, size_t align) {
-    off = (off + align - 1) & ~(align - 1);
-    off += bytes;
-  };
-  take(sizeof(double) * WG, 8);
-  take(sizeof(float) * k * kp, 4);
-  take(sizeof(float) * k * kp, 4);
-  take(sizeof(float) * k * dp, 4);
-  take(sizeof(float) * k, 4);
-  take(sizeof(float) * k, 4);
-  take(sizeof(float) * k, 4);
-  take(sizeof(float) * k, 4);
-  take(sizeof(float) * d, 4);
-  take(sizeof(int), 4);
+  off += sizeof(double) * 8 + sizeof(double) * 2;      // red + misc
+  off += sizeof(float) * (size_t)k * (k + 1);          // A
+  off += sizeof(float) * tsz;                          // T
+  off += sizeof(float) * (size_t)k * (d + 1);          // X
+  off += sizeof(float) * 4 * k;                        // yb alpha tvec rrow
+  off += sizeof(float) * d;                            // s2
+  off += sizeof(int) * 4;                              // bad (+pad)
   return off;
 }
 
-__device__ inline double block_reduce_sum(double v, double* red) {
-  const int tid = threadIdx.x;
-  red[tid] = v;
+__device__ inline NllLds carve(char* base, int k, int d) {
+  NllLds L;
+  size_t tsz = (size_t)(k * 33 > 32 * (k + 1) ? k * 33 : 32 * (k + 1));
+  char* p = base;
+  L.red = (double*)p;   p += sizeof(double) * 8;
+  L.misc = (double*)p;  p += sizeof(double) * 2;
+  L.A = (float*)p;      p += sizeof(float) * (size_t)k * (k + 1);
+  L.T = (float*)p;      p += sizeof(float) * tsz;
+  L.X = (float*)p;      p += sizeof(float) * (size_t)k * (d + 1);
+  L.yb = (float*)p;     p += sizeof(float) * k;
+  L.alpha = (float*)p;  p += sizeof(float) * k;
+  L.tvec = (float*)p;   p += sizeof(float) * k;
+  L.rrow = (float*)p;   p += sizeof(float) * k;
+  L.s2 = (float*)p;     p += sizeof(float) * d;
+  L.bad = (int*)p;
+  return L;
+}
+
+__device__ inline double wave_sum(double v) {
+#pragma unroll
+  for (int off = 32; off; off >>= 1) v += __shfl_down(v, off, 64);
+  return v;
+}
+
+__device__ inline double block_sum(double v, double* red, int tid) {
+  v = wave_sum(v);
+  if ((tid & 63) == 0) red[tid >> 6] = v;
   __syncthreads();
-  for (int s = WG / 2; s > 0; s >>= 1) {
-    if (tid < s) red[tid] += red[tid + s];
-    __syncthreads();
+  if (tid == 0) {
+    double s = 0;
+#pragma unroll
+    for (int w = 0; w < WG / 64; ++w) s += red[w];
+    red[0] = s;
   }
+  __syncthreads();
   double out = red[0];
   __syncthreads();
   return out;
+}
+
+// flat f -> (a, b) with a >= b in a lower triangle (incl. diagonal)
+__device__ inline void tri_decode(int f, int& a, int& b) {
+  a = (int)((sqrtf(8.f * (float)f + 1.f) - 1.f) * 0.5f);
+  while ((a + 1) * (a + 2) / 2 <= f) ++a;
+  while (a * (a + 1) / 2 > f) --a;
+  b = f - a * (a + 1) / 2;
 }
 
 extern "C" __global__ void __launch_bounds__(WG)
@@ -114,14 +128,18 @@ fused_expert_nll_kernel(const float* __restrict__ Xg,
                         double* __restrict__ out_contr,    // [E, d]
                         int* __restrict__ out_bad) {
   extern __shared__ char lds_raw[];
-  const int kp = k + 1, dp = d + 1;
-  NllLds S = carve_lds(lds_raw, k, d, kp, dp);
+  NllLds S = carve(lds_raw, k, d);
+  const int SA = k + 1;
+  const int dp = d + 1;
   const int e = blockIdx.x;
   const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int nblk = (k + NB - 1) / NB;
   const float* Xe = Xg + (size_t)e * k * d;
   const float* ye = yg + (size_t)e * k;
 
-  // ---- Phase A: stage X, y, s2 -------------------------------------
+  // ---- A: stage X, y, s2 -------------------------------------------
   for (int i = tid; i < k * d; i += WG) {
     int a = i / d, j = i - a * d;
     S.X[a * dp + j] = Xe[i];
@@ -131,56 +149,114 @@ fused_expert_nll_kernel(const float* __restrict__ Xg,
     float s = scale[j];
     S.s2[j] = s * s;
   }
-  if (tid == 0) *S.bad = 0;
+  if (tid == 0) { *S.bad = 0; S.misc[0] = 0.0; }
   __syncthreads();
 
-  // ---- Phase B: K = amp * exp(-q) + noise I ------------------------
-  // lower triangle computed, mirrored (k^2 d / WG work)
-  for (int f = tid; f < k * k; f += WG) {
-    int a = f / k, b = f - a * k;
-    if (b > a) continue;
-    float q = 0.f;
-    const float* xa = S.X + a * dp;
-    const float* xb = S.X + b * dp;
-    for (int j = 0; j < d; ++j) {
-      float t = xa[j] - xb[j];
-      q += S.s2[j] * t * t;
+  // ---- B: K lower = amp exp(-q) + noise I --------------------------
+  {
+    const int nlow = k * (k + 1) / 2;
+    for (int f = tid; f < nlow; f += WG) {
+      int a, b;
+      tri_decode(f, a, b);
+      float q = 0.f;
+      const float* xa = S.X + a * dp;
+      const float* xb = S.X + b * dp;
+      for (int j = 0; j < d; ++j) {
+        float t = xa[j] - xb[j];
+        q += S.s2[j] * t * t;
+      }
+      float kv = amp * __expf(-q);
+      if (a == b) kv += noise;
+      S.A[a * SA + b] = kv;
     }
-    float kv = amp * __expf(-q);
-    if (a == b) kv += noise;
-    S.A[a * kp + b] = kv;
-    S.A[b * kp + a] = kv;
   }
   __syncthreads();
 
-  // ---- Phase C: Cholesky (right-looking, fp32) ---------------------
-  for (int j = 0; j < k; ++j) {
-    const float ajj = S.A[j * kp + j];
-    if (!(ajj > 0.f) || !isfinite(ajj)) {      // uniform branch: all threads
-      if (tid == 0) *S.bad = 1;                // read the same LDS value
-      break;
+  // ---- C: blocked in-place Cholesky + diagonal-block inverse --------
+  for (int J = 0; J < nblk; ++J) {
+    const int jb = J * NB;
+    const int bs = min(NB, k - jb);
+    float* D = S.A + (size_t)jb * SA + jb;   // diag block, stride SA
+
+    if (wave == 0) {
+      double ldet = 0.0;
+      bool ok = true;
+      // potrf(bs) in place, lanes 0..bs-1 = rows
+      for (int s = 0; s < bs && ok; ++s) {
+        const float ajj = D[s * SA + s];
+        if (!(ajj > 0.f) || !isfinite(ajj)) { ok = false; break; }
+        ldet += (double)__logf(ajj);
+        const float rinv = rsqrtf(ajj);
+        __builtin_amdgcn_wave_barrier();
+        if (lane > s && lane < bs) D[lane * SA + s] *= rinv;
+        if (lane == s) D[s * SA + s] = ajj * rinv;   // sqrt(ajj)
+        __syncwarp();
+        if (lane > s && lane < bs) {
+          const float lis = D[lane * SA + s];
+          for (int c = s + 1; c <= lane; ++c)
+            D[lane * SA + c] -= lis * D[c * SA + s];
+        }
+        __syncwarp();
+      }
+      if (!ok) {
+        if (lane == 0) *S.bad = 1;
+      } else {
+        // trtri(bs) in place, row-stepping (row i reads original L row i
+        // and already-inverted rows < i)
+        for (int i = 0; i < bs; ++i) {
+          float v = 0.f;
+          if (lane < i) {
+            float s = 0.f;
+            for (int c = lane; c < i; ++c)
+              s += D[i * SA + c] * D[c * SA + lane];
+            v = -s / D[i * SA + i];
+          } else if (lane == i) {
+            v = 1.0f / D[i * SA + i];
+          }
+          __syncwarp();
+          if (lane <= i) D[i * SA + lane] = v;
+          __syncwarp();
+        }
+        if (lane == 0) S.misc[0] += ldet;
+      }
     }
-    const float rinv = rsqrtf(ajj);
     __syncthreads();
-    // scale column j; write the diagonal sqrt
-    for (int i = j + 1 + tid; i < k; i += WG) S.A[i * kp + j] *= rinv;
-    if (tid == 0) S.A[j * kp + j] = sqrtf(ajj);
-    __syncthreads();
-    // trailing rank-1 update of the lower triangle (rows/cols > j)
-    const int t = k - j - 1;                   // trailing size
-    const int n_el = t * (t + 1) / 2;
-    for (int f = tid; f < n_el; f += WG) {
-      // map flat f -> (ii >= cc) within trailing block
-      int ii = (int)((sqrtf(8.f * f + 1.f) - 1.f) * 0.5f);
-      while ((ii + 1) * (ii + 2) / 2 <= f) ++ii;   // fix fp rounding
-      while (ii * (ii + 1) / 2 > f) --ii;
-      int cc = f - ii * (ii + 1) / 2;
-      int i = j + 1 + ii, c = j + 1 + cc;
-      S.A[i * kp + c] -= S.A[i * kp + j] * S.A[c * kp + j];
+    if (*S.bad) break;
+
+    const int t0 = jb + bs;        // first trailing row
+    const int nr = k - t0;         // panel rows
+    if (nr > 0) {
+      // C2: copy panel below the diag block into T (T row r-t0, stride 33)
+      for (int f = tid; f < nr * bs; f += WG) {
+        int r = f / bs, c = f - r * bs;
+        S.T[r * 33 + c] = S.A[(size_t)(t0 + r) * SA + jb + c];
+      }
+      __syncthreads();
+      // C2b: panel <- T * V_JJ^T : A[r][jb+c] = sum_{t<=c} T[r][t] V[c][t]
+      for (int f = tid; f < nr * bs; f += WG) {
+        int r = f / bs, c = f - r * bs;
+        const float* Trow = S.T + r * 33;
+        const float* Vrow = D + c * SA;       // V_JJ row c (lower)
+        float s = 0.f;
+        for (int t = 0; t <= c; ++t) s += Trow[t] * Vrow[t];
+        S.A[(size_t)(t0 + r) * SA + jb + c] = s;
+      }
+      __syncthreads();
+      // C3: trailing update (lower incl. diag): A[i][c] -= L[i][Jb] . L[c][Jb]
+      const int ntri = nr * (nr + 1) / 2;
+      for (int f = tid; f < ntri; f += WG) {
+        int a, b;
+        tri_decode(f, a, b);
+        const int i = t0 + a, c = t0 + b;
+        const float* li = S.A + (size_t)i * SA + jb;
+        const float* lc = S.A + (size_t)c * SA + jb;
+        float s = 0.f;
+        for (int t = 0; t < bs; ++t) s += li[t] * lc[t];
+        S.A[(size_t)i * SA + c] -= s;
+      }
+      __syncthreads();
     }
-    __syncthreads();
   }
-  __syncthreads();
   if (*S.bad) {
     if (tid == 0) {
       out_bad[e] = 1;
@@ -189,100 +265,141 @@ fused_expert_nll_kernel(const float* __restrict__ Xg,
     for (int j = tid; j < d; j += WG) out_contr[(size_t)e * d + j] = 0.0;
     return;
   }
+  // log|K| = 2 sum log L_ii = sum log(ajj before sqrt), accumulated in C1
+  const double logdet = S.misc[0];
 
-  // ---- Phase C2: logdet = 2 sum log L_ii ---------------------------
-  double part = 0.0;
-  for (int i = tid; i < k; i += WG) part += log((double)S.A[i * kp + i]);
-  const double logdet = 2.0 * block_reduce_sum(part, S.red);
-
-  // ---- Phase D: V = L^-1 (one column per thread) -------------------
-  if (tid < k) {
-    const int j = tid;
-    S.V[j * kp + j] = 1.0f / S.A[j * kp + j];
-    for (int i = j + 1; i < k; ++i) {
+  // ---- D: off-diagonal triangular inverse, in place, J right-to-left
+  // V_IJ = -(sum_{K=J+1..I} V_IK L_KJ) L_JJ^-1 ; rows fully parallel.
+  for (int J = nblk - 2; J >= 0; --J) {
+    const int jb = J * NB;
+    const int bs = NB;                       // J < nblk-1 => full block
+    const int t0 = jb + bs;
+    const int nr = k - t0;
+    // U into T: U[r][t] = sum_{c=t0..t0+r} V[t0+r][c] * L[c][jb+t]
+    for (int f = tid; f < nr * bs; f += WG) {
+      int r = f / bs, t = f - r * bs;
+      const int row = t0 + r;
       float s = 0.f;
-      for (int c = j; c < i; ++c) s += S.A[i * kp + c] * S.V[c * kp + j];
-      S.V[i * kp + j] = -s / S.A[i * kp + i];
+      const float* vr = S.A + (size_t)row * SA;
+      for (int c = t0; c <= row; ++c)
+        s += vr[c] * S.A[(size_t)c * SA + jb + t];
+      S.T[r * 33 + t] = s;
     }
+    __syncthreads();
+    // V[row][jb+j] = - sum_{t>=j} U[r][t] * V_JJ[t][j]
+    for (int f = tid; f < nr * bs; f += WG) {
+      int r = f / bs, j = f - r * bs;
+      const float* Trow = S.T + r * 33;
+      float s = 0.f;
+      for (int t = j; t < bs; ++t)
+        s += Trow[t] * S.A[(size_t)(jb + t) * SA + jb + j];
+      S.A[(size_t)(t0 + r) * SA + jb + j] = -s;
+    }
+    __syncthreads();
   }
-  __syncthreads();
 
-  // ---- Phase E: alpha = V^T (V y); nll -----------------------------
+  // ---- E: alpha = V^T (V y), y.alpha ------------------------------
   for (int i = tid; i < k; i += WG) {
+    const float* vr = S.A + (size_t)i * SA;
     float s = 0.f;
-    for (int c = 0; c <= i; ++c) s += S.V[i * kp + c] * S.yb[c];
+    for (int c = 0; c <= i; ++c) s += vr[c] * S.yb[c];
     S.tvec[i] = s;
   }
   __syncthreads();
   for (int a = tid; a < k; a += WG) {
     float s = 0.f;
-    for (int i = a; i < k; ++i) s += S.V[i * kp + a] * S.tvec[i];
+    for (int i = a; i < k; ++i) s += S.A[(size_t)i * SA + a] * S.tvec[i];
     S.alpha[a] = s;
   }
   __syncthreads();
-  part = 0.0;
+  double part = 0.0;
   for (int i = tid; i < k; i += WG)
     part += (double)S.yb[i] * (double)S.alpha[i];
-  const double yta = block_reduce_sum(part, S.red);
+  const double yta = block_sum(part, S.red, tid);
 
-  // ---- Phase F: W0 = (aa^T - K^-1) o Kb into A; trG, sumW0 ---------
-  __syncthreads();            // everyone done reading L from A
-  double trg_part = 0.0, sw_part = 0.0;
-  const int n_low = k * (k + 1) / 2;
-  for (int f = tid; f < n_low; f += WG) {
-    int a = (int)((sqrtf(8.f * f + 1.f) - 1.f) * 0.5f);
-    while ((a + 1) * (a + 2) / 2 <= f) ++a;
-    while (a * (a + 1) / 2 > f) --a;
-    int b = f - a * (a + 1) / 2;          // b <= a
-    float kinv = 0.f;
-    for (int c = a; c < k; ++c) kinv += S.V[c * kp + a] * S.V[c * kp + b];
-    float g = S.alpha[a] * S.alpha[b] - kinv;
-    float q = 0.f;
-    const float* xa = S.X + a * dp;
-    const float* xb = S.X + b * dp;
-    for (int j = 0; j < d; ++j) {
-      float t = xa[j] - xb[j];
-      q += S.s2[j] * t * t;
+  // ---- L: K^-1 = V^T V in place (lauum), ascending row blocks ------
+  // element (i, j): sum_{c >= max(i,j)} V[c][i] V[c][j] — valid for any j.
+  for (int I = 0; I < nblk; ++I) {
+    const int ib = I * NB;
+    const int bs = min(NB, k - ib);
+    const int ncol = min(k, ib + bs);        // columns 0 .. ib+bs-1
+    for (int f = tid; f < bs * ncol; f += WG) {
+      const int r = f / ncol, j = f - r * ncol;
+      const int i = ib + r;
+      const int c0 = max(i, j);
+      float s = 0.f;
+      for (int c = c0; c < k; ++c)
+        s += S.A[(size_t)c * SA + i] * S.A[(size_t)c * SA + j];
+      S.T[r * (k + 1) + j] = s;
     }
-    float w = g * __expf(-q);
-    if (a == b) {
-      trg_part += (double)g;
-      sw_part += (double)w;
-    } else {
-      sw_part += 2.0 * (double)w;
+    __syncthreads();
+    for (int f = tid; f < bs * ncol; f += WG) {
+      const int r = f / ncol, j = f - r * ncol;
+      S.A[(size_t)(ib + r) * SA + j] = S.T[r * (k + 1) + j];
     }
-    S.A[a * kp + b] = w;
-    S.A[b * kp + a] = w;
+    __syncthreads();
   }
-  const double trG = block_reduce_sum(trg_part, S.red);
-  const double sumW0 = block_reduce_sum(sw_part, S.red);
 
-  // ---- Phase G: row sums of W0 -------------------------------------
+  // ---- W: W0 = (aa^T - K^-1) o Kb, in place + mirror; trG, sumW0 ---
+  double trg_part = 0.0, sw_part = 0.0;
+  {
+    const int nlow = k * (k + 1) / 2;
+    for (int f = tid; f < nlow; f += WG) {
+      int a, b;
+      tri_decode(f, a, b);
+      const float g = S.alpha[a] * S.alpha[b] - S.A[(size_t)a * SA + b];
+      float q = 0.f;
+      const float* xa = S.X + a * dp;
+      const float* xb = S.X + b * dp;
+      for (int j = 0; j < d; ++j) {
+        float t = xa[j] - xb[j];
+        q += S.s2[j] * t * t;
+      }
+      const float w = g * __expf(-q);
+      if (a == b) {
+        trg_part += (double)g;
+        sw_part += (double)w;
+      } else {
+        sw_part += 2.0 * (double)w;
+      }
+      S.A[(size_t)a * SA + b] = w;
+      S.A[(size_t)b * SA + a] = w;
+    }
+  }
+  const double trG = block_sum(trg_part, S.red, tid);
+  const double sumW0 = block_sum(sw_part, S.red, tid);
+
+  // ---- G: row sums of W0 ------------------------------------------
   for (int a = tid; a < k; a += WG) {
+    const float* wr = S.A + (size_t)a * SA;
     float s = 0.f;
-    for (int b = 0; b < k; ++b) s += S.A[a * kp + b];
+    for (int b = 0; b < k; ++b) s += wr[b];
     S.rrow[a] = s;
   }
   __syncthreads();
 
-  // ---- Phase H: WX = W0 @ X into V (d <= k columns fit) ------------
-  for (int f = tid; f < k * d; f += WG) {
-    int a = f / d, j = f - a * d;
-    float s = 0.f;
-    for (int b = 0; b < k; ++b) s += S.A[a * kp + b] * S.X[b * dp + j];
-    S.V[a * kp + j] = s;
-  }
-  __syncthreads();
-
-  // contr_j = 2 sum_a x_aj^2 r_a - 2 sum_a x_aj WX_aj   (fp64 accum)
-  for (int j = tid; j < d; j += WG) {
-    double acc = 0.0;
-    for (int a = 0; a < k; ++a) {
-      float x = S.X[a * dp + j];
-      acc += 2.0 * (double)x *
-             ((double)x * (double)S.rrow[a] - (double)S.V[a * kp + j]);
+  // ---- H: WX = W0 @ X (d-halves through T); contraction ------------
+  for (int d0 = 0; d0 < d; d0 += 32) {
+    const int dl = min(32, d - d0);
+    for (int f = tid; f < k * dl; f += WG) {
+      const int a = f / dl, j = f - a * dl;
+      const float* wr = S.A + (size_t)a * SA;
+      float s = 0.f;
+      for (int b = 0; b < k; ++b) s += wr[b] * S.X[b * dp + d0 + j];
+      S.T[a * 33 + j] = s;
     }
-    out_contr[(size_t)e * d + j] = acc;
+    __syncthreads();
+    // contr_j = 2 sum_a x_aj^2 r_a - 2 sum_a x_aj WX_aj   (fp64)
+    for (int j = tid; j < dl; j += WG) {
+      double acc = 0.0;
+      for (int a = 0; a < k; ++a) {
+        const float x = S.X[a * dp + d0 + j];
+        acc += 2.0 * (double)x *
+               ((double)x * (double)S.rrow[a] - (double)S.T[a * 33 + j]);
+      }
+      out_contr[(size_t)e * d + d0 + j] = acc;
+    }
+    __syncthreads();
   }
 
   if (tid == 0) {
@@ -299,10 +416,10 @@ extern "C" hipError_t launch_fused_expert_nll(
     float amp, float noise, int E, int k, int d,
     double* out_nll, double* out_sumW0, double* out_trG, double* out_contr,
     int* out_bad, hipStream_t stream, size_t* lds_used) {
-  const int kp = k + 1, dp = d + 1;
-  size_t lds = nll_lds_bytes(k, d, kp, dp);
+  size_t lds = nll_lds_bytes2(k, d);
   if (lds_used) *lds_used = lds;
-  if (lds > 160 * 1024) return hipErrorInvalidConfiguration;
+  if (lds > 160 * 1024 || k > 128 || d > k || d > 64)
+    return hipErrorInvalidConfiguration;
   hipLaunchKernelGGL(fused_expert_nll_kernel, dim3(E), dim3(WG), lds, stream,
                      X, y, scale, amp, noise, k, d,
                      out_nll, out_sumW0, out_trG, out_contr, out_bad);

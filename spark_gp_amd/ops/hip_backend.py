@@ -97,7 +97,7 @@ def _s2_vector(cs: CompiledKernel, theta: np.ndarray, d: int, device):
 
 def kmn_knm_and_kmny(kernel: Kernel, active: torch.Tensor,
                      X: torch.Tensor, y: torch.Tensor,
-                     chunk_rows: int = 131072
+                     chunk_rows: int = 131072, precision: str = "fp64"
                      ) -> Tuple[torch.Tensor, torch.Tensor]:
     cs = compile_kernel(kernel)
     theta = kernel.get_hyperparameters()
@@ -108,16 +108,29 @@ def kmn_knm_and_kmny(kernel: Kernel, active: torch.Tensor,
     act32 = active.to(torch.float32).contiguous()
     y32 = y.to(torch.float32)
 
-    KK = torch.zeros(m, m, dtype=torch.float32, device=X.device)
     Ky = torch.zeros(m, dtype=torch.float64, device=X.device)
+    if precision == "fp64":
+        # reference-parity numerics: K_nm values from the HIP cross kernel
+        # (fp32, ~6e-8 quantization), products and accumulation in fp64
+        KK64 = torch.zeros(m, m, dtype=torch.float64, device=X.device)
+        for s in range(0, n, chunk_rows):
+            e = min(n, s + chunk_rows)
+            Kc = ext.cross_kernel_tile(X[s:e].contiguous(), act32, s2,
+                                       float(C), False, False)[0]
+            K64 = Kc.double()
+            KK64 += K64.transpose(0, 1) @ K64
+            Ky += K64.transpose(0, 1) @ y[s:e].double()
+        return KK64, Ky
+
+    # 'mixed': hi/lo bf16 split — KK += hi^T hi + hi^T lo + lo^T hi on the
+    # MFMA SYRK keeps input-quantization error at fp32 class while running
+    # at bf16 matrix-core rate
+    KK = torch.zeros(m, m, dtype=torch.float32, device=X.device)
     ntile = (m + 127) // 128
     tiles = ntile * (ntile + 1) // 2
     split_k = max(1, min(64, (512 + tiles - 1) // tiles))
     for s in range(0, n, chunk_rows):
         e = min(n, s + chunk_rows)
-        # hi/lo bf16 split: KK += hi^T hi + hi^T lo + lo^T hi keeps the
-        # input-quantization error at fp32 level so the PD matrix's small
-        # eigenvalues (~sigma2 * lambda_min(Kmm)) survive
         Kc, Kl = ext.cross_kernel_tile(X[s:e].contiguous(), act32, s2,
                                        float(C), True, True)
         ext.syrk_bf16_acc(Kc, Kl, KK, split_k)

@@ -34,8 +34,9 @@ class NotPositiveDefiniteError(RuntimeError):
 
 def accumulate_ppa_stats(kernel: Kernel, active: torch.Tensor,
                          X: torch.Tensor, y: torch.Tensor,
-                         comm: Comm) -> Tuple[torch.Tensor, torch.Tensor]:
-    KK, Ky = ops.kmn_knm_and_kmny(kernel, active, X, y)
+                         comm: Comm, precision: str = "fp64"
+                         ) -> Tuple[torch.Tensor, torch.Tensor]:
+    KK, Ky = ops.kmn_knm_and_kmny(kernel, active, X, y, precision=precision)
     comm.allreduce_(KK)
     comm.allreduce_(Ky)
     return KK, Ky

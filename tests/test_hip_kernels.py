@@ -200,6 +200,28 @@ def test_gpu_fit_end_to_end(dev, ext):
     assert err < 0.02, f"GPU fit rmse {err}"
 
 
+def test_gpu_fit_mixed_precision_ppa(dev, ext):
+    """Same fit with the hi/lo bf16 MFMA SYRK path: fastest configuration,
+    quality within a looser band."""
+    from spark_gp_amd import GaussianProcessRegression, rmse
+    from spark_gp_amd.kernels import ARDRBFKernel
+    rng = np.random.default_rng(0)
+    X = rng.random((30000, 4)).astype(np.float32)
+    y = (np.sin(5.0 * X[:, 0]) + X[:, 1]).astype(np.float32)
+    gp = (GaussianProcessRegression()
+          .setKernel(lambda: 1 * ARDRBFKernel(4))
+          .setDatasetSizeForExpert(100)
+          .setActiveSetSize(300)
+          .setSigma2(1e-3)
+          .setMaxIter(15)
+          .setSeed(3)
+          .setPpaPrecision("mixed")
+          .setDevice("cuda:0"))
+    model = gp.fit(X, y)
+    err = rmse(y[:3000], model.predict(X[:3000]))
+    assert err < 0.1, f"mixed-precision GPU fit rmse {err}"
+
+
 def test_gpu_classifier_end_to_end(dev, ext):
     import os
     from spark_gp_amd import GaussianProcessClassifier, RBFKernel, accuracy
