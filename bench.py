@@ -45,6 +45,9 @@ def parse_args():
     p.add_argument("--ppa-precision", type=str, default="mixed",
                    choices=["mixed", "fp64"],
                    help="PPA SYRK path: hi/lo bf16 MFMA (mixed) or fp64")
+    p.add_argument("--min-warmup-seconds", type=float, default=8.0,
+                   help="keep running warmup fits until this much wall time "
+                        "has passed (DVFS clock stabilization)")
     return p.parse_args()
 
 
@@ -87,8 +90,16 @@ def main():
             torch.cuda.synchronize()
 
     model = None
-    for _ in range(args.warmup):
+    # Warmup: at least the requested fits, AND enough sustained GPU load for
+    # DVFS to reach steady-state clocks (MI355X ramps over seconds; timing
+    # the first fits on a cold GPU under-reports steady throughput by 3-4x).
+    t_w = time.perf_counter()
+    w = 0
+    while w < args.warmup or (device.type == "cuda"
+                              and time.perf_counter() - t_w
+                              < args.min_warmup_seconds):
         model = make_gp().fit(Xt, yt)
+        w += 1
 
     sync()
     t0 = time.perf_counter()

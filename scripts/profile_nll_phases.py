@@ -1,0 +1,63 @@
+"""Per-phase timing of the fused expert NLL kernel (wall_clock64, 100 MHz)."""
+import os
+import sys
+import time
+
+import numpy as np
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+from spark_gp_amd import _hip_ext as ext
+
+E, k, d = 20000, 100, 32
+g = torch.Generator().manual_seed(0)
+X = torch.rand(E, k, d, generator=g).cuda()
+y = torch.sin(3 * X.sum(-1)).cuda()
+scale = torch.rand(d, generator=g).add(0.5).cuda()
+
+# warmup + timing
+for _ in range(2):
+    out = ext.fused_expert_nll(X, y, scale, 1.0, 1e-3)
+torch.cuda.synchronize()
+t0 = time.perf_counter()
+for _ in range(5):
+    out = ext.fused_expert_nll(X, y, scale, 1.0, 1e-3)
+torch.cuda.synchronize()
+dt = (time.perf_counter() - t0) / 5
+print(f"E={E} k={k} d={d}: {dt*1e3:.2f} ms/launch, "
+      f"{dt/E*1e6:.2f} us/expert-slot")
+
+*_, clk = ext.fused_expert_nll_profile(X, y, scale, 1.0, 1e-3, True)
+clk = clk.cpu().numpy().astype(np.float64)
+names = ["stage", "B build", "C chol", "D trtri", "E alpha", "L lauum",
+         "W w0", "G rowsum", "H contr", "out"]
+# per-expert deltas in microseconds (wall_clock64 = 100 MHz)
+FREQ = 1e8
+deltas = (clk[:, 1:10] - clk[:, 0:9]) / FREQ * 1e6
+mean = deltas.mean(0)
+total = mean.sum()
+print(f"sum of phases: {total:.2f} us (mean per expert)")
+for i, n in enumerate(names[:9]):
+    print(f"  {names[i+1] if False else n:>8} -> {names[i+1]:>8}: "
+          f"{mean[i]:8.2f} us  ({100*mean[i]/total:4.1f}%)")
+c1 = clk[:, 10].mean() / FREQ * 1e6
+print(f"C1 (diag factor+inv) within C: {c1:.2f} us")
+# wall span of the whole launch from clocks
+span = (clk[:, 9].max() - clk[:, 0].min()) / FREQ * 1e3
+print(f"launch span by clocks: {span:.2f} ms")
+
+# contention probe: few blocks (1 per CU) vs full load
+for Ee in (256, 2048):
+    Xs, ys = X[:Ee], y[:Ee]
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(10):
+        ext.fused_expert_nll(Xs, ys, scale, 1.0, 1e-3)
+    torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / 10
+    *_, clk2 = ext.fused_expert_nll_profile(Xs, ys, scale, 1.0, 1e-3, True)
+    clk2 = clk2.cpu().numpy().astype(np.float64)
+    dd = (clk2[:, 1:10] - clk2[:, 0:9]) / 1e8 * 1e6
+    m = dd.mean(0)
+    print(f"E={Ee}: {dt*1e3:.3f} ms/launch; phases us: "
+          + " ".join(f"{names[i+1] if i<9 else ''}:{m[i]:.1f}" for i in range(9)))

@@ -9,8 +9,8 @@
 extern "C" hipError_t launch_fused_expert_nll(
     const float* X, const float* y, const float* scale, float amp,
     float noise, int E, int k, int d, double* out_nll, double* out_sumW0,
-    double* out_trG, double* out_contr, int* out_bad, hipStream_t stream,
-    size_t* lds_used);
+    double* out_trG, double* out_contr, int* out_bad,
+    unsigned long long* out_clk, hipStream_t stream, size_t* lds_used);
 
 extern "C" hipError_t launch_cross_kernel_tile(
     const float* X, const float* A, const float* s2v, float amp, int c,
@@ -38,9 +38,11 @@ hipStream_t current_stream() {
 }  // namespace
 
 // Returns (nll[E] f64, sumW0[E] f64, trG[E] f64, contr[E,d] f64, bad[E] i32)
-std::vector<torch::Tensor> fused_expert_nll(torch::Tensor X, torch::Tensor y,
-                                            torch::Tensor scale, double amp,
-                                            double noise) {
+std::vector<torch::Tensor> fused_expert_nll_impl(torch::Tensor X,
+                                                 torch::Tensor y,
+                                                 torch::Tensor scale,
+                                                 double amp, double noise,
+                                                 bool profile) {
   TORCH_CHECK(X.is_cuda() && X.dtype() == torch::kFloat32 && X.dim() == 3,
               "X must be [E, k, d] float32 on GPU");
   TORCH_CHECK(y.is_cuda() && y.dtype() == torch::kFloat32 && y.dim() == 2);
@@ -57,15 +59,30 @@ std::vector<torch::Tensor> fused_expert_nll(torch::Tensor X, torch::Tensor y,
   auto trG = torch::empty({E}, opts64);
   auto contr = torch::empty({E, d}, opts64);
   auto bad = torch::empty({E}, opts32i);
+  torch::Tensor clk;
+  unsigned long long* clk_ptr = nullptr;
+  if (profile) {
+    clk = torch::zeros({E, 12},
+                       torch::TensorOptions().dtype(torch::kInt64)
+                           .device(X.device()));
+    clk_ptr = (unsigned long long*)clk.data_ptr<int64_t>();
+  }
   size_t lds = 0;
   check_hip(launch_fused_expert_nll(
                 Xc.data_ptr<float>(), yc.data_ptr<float>(),
                 sc.data_ptr<float>(), (float)amp, (float)noise, E, k, d,
                 nll.data_ptr<double>(), sumW0.data_ptr<double>(),
                 trG.data_ptr<double>(), contr.data_ptr<double>(),
-                bad.data_ptr<int>(), current_stream(), &lds),
+                bad.data_ptr<int>(), clk_ptr, current_stream(), &lds),
             "fused_expert_nll");
+  if (profile) return {nll, sumW0, trG, contr, bad, clk};
   return {nll, sumW0, trG, contr, bad};
+}
+
+std::vector<torch::Tensor> fused_expert_nll(torch::Tensor X, torch::Tensor y,
+                                            torch::Tensor scale, double amp,
+                                            double noise) {
+  return fused_expert_nll_impl(X, y, scale, amp, noise, false);
 }
 
 bool fused_expert_nll_supported(int64_t k, int64_t d) {
@@ -137,6 +154,8 @@ void colsum_gemv_acc(torch::Tensor Kc, torch::Tensor y, torch::Tensor Ky) {
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("fused_expert_nll", &fused_expert_nll,
           "fused per-expert BCM nll+gradient primitives (CDNA4)");
+  mod.def("fused_expert_nll_profile", &fused_expert_nll_impl,
+          "same, with per-phase wall_clock64 boundaries appended");
   mod.def("fused_expert_nll_supported", &fused_expert_nll_supported);
   mod.def("cross_kernel_tile", &cross_kernel_tile,
           "rectangular RBF/ARD kernel block (CDNA4)");

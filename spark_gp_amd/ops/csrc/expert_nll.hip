@@ -8,25 +8,26 @@
 // 48-79 — K and d derivative matrices) executes in LDS without touching HBM
 // between stages, and WITHOUT materializing the [d, k, k] derivative tensor:
 //
-//   B  K = amp * exp(-sum_d s2_d (x_ad - x_bd)^2) + noise I   (lower, LDS)
+//   B  lower(A) = amp*Kb + noise*I,  strict upper(A) = Kb (cached base
+//      kernel — read back in phase W, never recomputed)
 //   C  blocked right-looking Cholesky, IN PLACE: per 32-column block, the
 //      diagonal block is factored AND inverted by one wave (lockstep
-//      __syncwarp steps, no workgroup barriers), the panel solve becomes a
-//      dense GEMM against the inverted diagonal, the trailing update is an
-//      all-thread data-parallel GEMM.  fp32; fp64 logdet on the fly.
+//      __syncwarp steps, both 32-lane halves cooperating), the panel solve
+//      becomes a dense GEMM against the inverted diagonal, the trailing
+//      update is an all-thread data-parallel GEMM.  fp32; fp64 logdet.
 //   D  blocked in-place triangular inverse of the off-diagonal blocks
-//      (right-to-left column blocks; every row is independent, fully
-//      parallel — the identity V_IJ = -(sum_{K>J} V_IK L_KJ) L_JJ^-1).
+//      (right-to-left column blocks, rows fully parallel:
+//       V_IJ = -(sum_{K>J} V_IK L_KJ) L_JJ^-1).
 //   E  alpha = V^T (V y);  nll = 1/2 y.alpha + 1/2 logdet
 //   L  in-place lauum: K^-1 = V^T V (ascending row blocks through a temp
-//      row buffer)
-//   W  W0 = (alpha alpha^T - K^-1) o Kb, Kb recomputed elementwise from X;
+//      row buffer; strictly lower+diagonal so the Kb cache survives)
+//   W  W0 = (alpha alpha^T - K^-1) o Kb (Kb from the upper-triangle cache);
 //      trG / sum(W0) accumulated on the fly
 //   H  gradient contraction per input dim (the K5 fusion):
 //      contr_j = 2 sum_a x_aj^2 r_a - 2 sum_a x_aj (W0 X)_aj
 //
-// Host-side chain rule turns (contr, sumW0, trG) into the gradient for both
-// ARD-RBF and RBF (ops/hip_backend.py).  Single k x (k+1) working buffer +
+// All serial inner products are multi-accumulator-unrolled so they run at
+// LDS throughput instead of FMA-latency.  Single k x (k+1) working buffer +
 // one k x 33 temp => ~69 KB LDS at k=100, two experts resident per CU.
 //
 // Numerics: fp32 storage/factorization, fp64 scalar accumulation.  Experts
@@ -42,7 +43,7 @@
 #define NB 32
 
 struct NllLds {
-  float* A;     // k * (k+1)  K -> L/V -> K^-1 -> W0
+  float* A;     // k * (k+1):  lower K -> L/V -> K^-1 -> W0; upper: Kb cache
   float* T;     // temp: max(k*33, 32*(k+1))
   float* X;     // k * (d+1) raw features
   float* yb;    // k
@@ -116,7 +117,22 @@ __device__ inline void tri_decode(int f, int& a, int& b) {
   b = f - a * (a + 1) / 2;
 }
 
-extern "C" __global__ void __launch_bounds__(WG)
+// 4-accumulator strided dot over LDS: sum_{c=c0}^{c1-1} p[c*sp] * q[c*sq]
+__device__ inline float dot4(const float* p, int sp, const float* q, int sq,
+                             int c0, int c1) {
+  float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
+  int c = c0;
+  for (; c + 3 < c1; c += 4) {
+    s0 += p[c * sp] * q[c * sq];
+    s1 += p[(c + 1) * sp] * q[(c + 1) * sq];
+    s2 += p[(c + 2) * sp] * q[(c + 2) * sq];
+    s3 += p[(c + 3) * sp] * q[(c + 3) * sq];
+  }
+  for (; c < c1; ++c) s0 += p[c * sp] * q[c * sq];
+  return (s0 + s1) + (s2 + s3);
+}
+
+extern "C" __global__ void __launch_bounds__(WG, 4)
 fused_expert_nll_kernel(const float* __restrict__ Xg,
                         const float* __restrict__ yg,
                         const float* __restrict__ scale,   // [d]
@@ -126,15 +142,19 @@ fused_expert_nll_kernel(const float* __restrict__ Xg,
                         double* __restrict__ out_sumW0,    // [E]
                         double* __restrict__ out_trG,      // [E]
                         double* __restrict__ out_contr,    // [E, d]
-                        int* __restrict__ out_bad) {
+                        int* __restrict__ out_bad,
+                        unsigned long long* __restrict__ out_clk) { // [E,12]
+                        // optional phase profiling (wall_clock64 boundaries)
   extern __shared__ char lds_raw[];
+#define PH(n) do { if (out_clk && threadIdx.x == 0) \
+    out_clk[(size_t)blockIdx.x * 12 + (n)] = wall_clock64(); } while (0)
+  PH(0);
   NllLds S = carve(lds_raw, k, d);
   const int SA = k + 1;
   const int dp = d + 1;
   const int e = blockIdx.x;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
-  const int wave = tid >> 6;
   const int nblk = (k + NB - 1) / NB;
   const float* Xe = Xg + (size_t)e * k * d;
   const float* ye = yg + (size_t)e * k;
@@ -152,81 +172,126 @@ fused_expert_nll_kernel(const float* __restrict__ Xg,
   if (tid == 0) { *S.bad = 0; S.misc[0] = 0.0; }
   __syncthreads();
 
-  // ---- B: K lower = amp exp(-q) + noise I --------------------------
+PH(1);
+    // ---- B: lower = amp Kb + noise I; strict upper = Kb cache --------
   {
     const int nlow = k * (k + 1) / 2;
     for (int f = tid; f < nlow; f += WG) {
       int a, b;
       tri_decode(f, a, b);
-      float q = 0.f;
       const float* xa = S.X + a * dp;
       const float* xb = S.X + b * dp;
-      for (int j = 0; j < d; ++j) {
-        float t = xa[j] - xb[j];
-        q += S.s2[j] * t * t;
+      float q0 = 0.f, q1 = 0.f;
+      int j = 0;
+      for (; j + 1 < d; j += 2) {
+        float t0 = xa[j] - xb[j];
+        float t1 = xa[j + 1] - xb[j + 1];
+        q0 += S.s2[j] * t0 * t0;
+        q1 += S.s2[j + 1] * t1 * t1;
       }
-      float kv = amp * __expf(-q);
-      if (a == b) kv += noise;
-      S.A[a * SA + b] = kv;
+      if (j < d) { float t = xa[j] - xb[j]; q0 += S.s2[j] * t * t; }
+      const float kb = __expf(-(q0 + q1));
+      S.A[a * SA + b] = amp * kb + (a == b ? noise : 0.f);
+      if (a != b) S.A[b * SA + a] = kb;        // Kb cache in the upper
     }
   }
   __syncthreads();
 
-  // ---- C: blocked in-place Cholesky + diagonal-block inverse --------
+PH(2);
+    // ---- C: blocked in-place Cholesky + diagonal-block inverse --------
   for (int J = 0; J < nblk; ++J) {
     const int jb = J * NB;
     const int bs = min(NB, k - jb);
     float* D = S.A + (size_t)jb * SA + jb;   // diag block, stride SA
 
-    if (wave == 0) {
-      double ldet = 0.0;
-      bool ok = true;
-      // potrf(bs) in place, lanes 0..bs-1 = rows
-      for (int s = 0; s < bs && ok; ++s) {
-        const float ajj = D[s * SA + s];
-        if (!(ajj > 0.f) || !isfinite(ajj)) { ok = false; break; }
-        ldet += (double)__logf(ajj);
-        const float rinv = rsqrtf(ajj);
-        __builtin_amdgcn_wave_barrier();
-        if (lane > s && lane < bs) D[lane * SA + s] *= rinv;
-        if (lane == s) D[s * SA + s] = ajj * rinv;   // sqrt(ajj)
-        __syncwarp();
-        if (lane > s && lane < bs) {
-          const float lis = D[lane * SA + s];
-          for (int c = s + 1; c <= lane; ++c)
-            D[lane * SA + c] -= lis * D[c * SA + s];
-        }
-        __syncwarp();
-      }
-      if (!ok) {
-        if (lane == 0) *S.bad = 1;
-      } else {
-        // trtri(bs) in place, row-stepping (row i reads original L row i
-        // and already-inverted rows < i)
-        for (int i = 0; i < bs; ++i) {
-          float v = 0.f;
-          if (lane < i) {
-            float s = 0.f;
-            for (int c = lane; c < i; ++c)
-              s += D[i * SA + c] * D[c * SA + lane];
-            v = -s / D[i * SA + i];
-          } else if (lane == i) {
-            v = 1.0f / D[i * SA + i];
+    unsigned long long c1t0 = 0;
+    if (out_clk && tid == 0) c1t0 = wall_clock64();
+    if (tid < 64) {
+      // wave 0: register-resident potrf + trtri of the bs x bs diagonal
+      // block, fully unrolled so every register-array index is static and
+      // the whole factorization runs on VGPRs + cross-lane shfl — no LDS
+      // round-trips in the serial dependency chain.
+      const int row = lane & 31;             // both 32-lane halves mirror
+      const int half = lane >> 5;
+      {
+        // potrf with deferred column scaling: the trailing update works on
+        // RAW columns (A[i][c] -= A[i][s] A[c][s] / a_ss), so each step's
+        // only cross-step LDS dependency is the pivot read -> ~2 LDS hops
+        // per step instead of 4.  Columns are scaled to L in one parallel
+        // pass at the end (pivots survive: column s is final at step s).
+        double ldet = 0.0;
+        bool ok = true;
+        for (int ss = 0; ss < bs; ++ss) {
+          const float ajj = D[ss * SA + ss];   // broadcast LDS read
+          if (!(ajj > 0.f) || !isfinite(ajj)) { ok = false; break; }
+          ldet += (double)__logf(ajj);
+          const float inv = __builtin_amdgcn_rcpf(ajj);
+          if (row > ss && row < bs) {
+            const float ris = D[row * SA + ss] * inv;
+            // trailing rank-1 update; the two 32-lane halves split by
+            // parity; 4-wide batches so loads share one lgkm wait
+            float* wr = D + (size_t)row * SA;
+            const float* pc = D + ss;
+            int c = ss + 1 + half;
+            for (; c + 6 <= row; c += 8) {
+              const float p0 = pc[c * SA], p1 = pc[(c + 2) * SA];
+              const float p2 = pc[(c + 4) * SA], p3 = pc[(c + 6) * SA];
+              const float a0 = wr[c], a1 = wr[c + 2];
+              const float a2 = wr[c + 4], a3 = wr[c + 6];
+              wr[c] = a0 - ris * p0;
+              wr[c + 2] = a1 - ris * p1;
+              wr[c + 4] = a2 - ris * p2;
+              wr[c + 6] = a3 - ris * p3;
+            }
+            for (; c <= row; c += 2)
+              wr[c] -= ris * pc[c * SA];
           }
           __syncwarp();
-          if (lane <= i) D[i * SA + lane] = v;
+        }
+        if (ok) {
+          // scale pass: L[i][s] = raw[i][s] * rsqrt(raw[s][s]) (diag incl.)
+          if (lane < bs) S.T[lane] = rsqrtf(D[lane * SA + lane]);
+          __syncwarp();
+          const int ntri = bs * (bs + 1) / 2;
+          for (int f = lane; f < ntri; f += 64) {
+            int i, c;
+            tri_decode(f, i, c);
+            D[i * SA + c] *= S.T[c];
+          }
           __syncwarp();
         }
-        if (lane == 0) S.misc[0] += ldet;
+        if (lane == 0) {
+          if (ok) S.misc[0] += ldet;
+          else *S.bad = 1;
+        }
+      }
+      __syncwarp();
+      if (!*S.bad) {
+        // trtri in place: step i reads L row i (broadcast) and the V
+        // entries of earlier steps (already written into D), then
+        // overwrites row i with V row i.  Wave lockstep orders the
+        // within-step reads before the write.
+        for (int i = 0; i < bs; ++i) {
+          const float* Li = D + i * SA;
+          const float sacc = dot4(Li, 1, D + row, SA, row, i);
+          const float rli = __builtin_amdgcn_rcpf(Li[i]);
+          const float vi = (row < i) ? (-sacc * rli)
+                                     : (row == i ? rli : 0.f);
+          __syncwarp();
+          if (lane < 32 && row <= i) D[i * SA + row] = vi;
+          __syncwarp();
+        }
       }
     }
+    if (out_clk && tid == 0)
+      out_clk[(size_t)blockIdx.x * 12 + 10] += wall_clock64() - c1t0;
     __syncthreads();
     if (*S.bad) break;
 
     const int t0 = jb + bs;        // first trailing row
     const int nr = k - t0;         // panel rows
     if (nr > 0) {
-      // C2: copy panel below the diag block into T (T row r-t0, stride 33)
+      // C2: copy panel below the diag block into T (row r-t0, stride 33)
       for (int f = tid; f < nr * bs; f += WG) {
         int r = f / bs, c = f - r * bs;
         S.T[r * 33 + c] = S.A[(size_t)(t0 + r) * SA + jb + c];
@@ -235,24 +300,19 @@ fused_expert_nll_kernel(const float* __restrict__ Xg,
       // C2b: panel <- T * V_JJ^T : A[r][jb+c] = sum_{t<=c} T[r][t] V[c][t]
       for (int f = tid; f < nr * bs; f += WG) {
         int r = f / bs, c = f - r * bs;
-        const float* Trow = S.T + r * 33;
-        const float* Vrow = D + c * SA;       // V_JJ row c (lower)
-        float s = 0.f;
-        for (int t = 0; t <= c; ++t) s += Trow[t] * Vrow[t];
-        S.A[(size_t)(t0 + r) * SA + jb + c] = s;
+        S.A[(size_t)(t0 + r) * SA + jb + c] =
+            dot4(S.T + r * 33, 1, D + c * SA, 1, 0, c + 1);
       }
       __syncthreads();
-      // C3: trailing update (lower incl. diag): A[i][c] -= L[i][Jb] . L[c][Jb]
+      // C3: trailing update (lower incl. diag): A[i][c] -= L[i][Jb].L[c][Jb]
       const int ntri = nr * (nr + 1) / 2;
       for (int f = tid; f < ntri; f += WG) {
         int a, b;
         tri_decode(f, a, b);
         const int i = t0 + a, c = t0 + b;
-        const float* li = S.A + (size_t)i * SA + jb;
-        const float* lc = S.A + (size_t)c * SA + jb;
-        float s = 0.f;
-        for (int t = 0; t < bs; ++t) s += li[t] * lc[t];
-        S.A[(size_t)i * SA + c] -= s;
+        S.A[(size_t)i * SA + c] -=
+            dot4(S.A + (size_t)i * SA + jb, 1, S.A + (size_t)c * SA + jb, 1,
+                 0, bs);
       }
       __syncthreads();
     }
@@ -268,7 +328,8 @@ fused_expert_nll_kernel(const float* __restrict__ Xg,
   // log|K| = 2 sum log L_ii = sum log(ajj before sqrt), accumulated in C1
   const double logdet = S.misc[0];
 
-  // ---- D: off-diagonal triangular inverse, in place, J right-to-left
+PH(3);
+    // ---- D: off-diagonal triangular inverse, in place, J right-to-left
   // V_IJ = -(sum_{K=J+1..I} V_IK L_KJ) L_JJ^-1 ; rows fully parallel.
   for (int J = nblk - 2; J >= 0; --J) {
     const int jb = J * NB;
@@ -279,68 +340,56 @@ fused_expert_nll_kernel(const float* __restrict__ Xg,
     for (int f = tid; f < nr * bs; f += WG) {
       int r = f / bs, t = f - r * bs;
       const int row = t0 + r;
-      float s = 0.f;
-      const float* vr = S.A + (size_t)row * SA;
-      for (int c = t0; c <= row; ++c)
-        s += vr[c] * S.A[(size_t)c * SA + jb + t];
-      S.T[r * 33 + t] = s;
+      S.T[r * 33 + t] = dot4(S.A + (size_t)row * SA, 1,
+                             S.A + jb + t, SA, t0, row + 1);
     }
     __syncthreads();
     // V[row][jb+j] = - sum_{t>=j} U[r][t] * V_JJ[t][j]
     for (int f = tid; f < nr * bs; f += WG) {
       int r = f / bs, j = f - r * bs;
-      const float* Trow = S.T + r * 33;
-      float s = 0.f;
-      for (int t = j; t < bs; ++t)
-        s += Trow[t] * S.A[(size_t)(jb + t) * SA + jb + j];
-      S.A[(size_t)(t0 + r) * SA + jb + j] = -s;
+      S.A[(size_t)(t0 + r) * SA + jb + j] =
+          -dot4(S.T + r * 33, 1, S.A + (size_t)jb * SA + jb + j, SA, j, bs);
     }
     __syncthreads();
   }
 
-  // ---- E: alpha = V^T (V y), y.alpha ------------------------------
-  for (int i = tid; i < k; i += WG) {
-    const float* vr = S.A + (size_t)i * SA;
-    float s = 0.f;
-    for (int c = 0; c <= i; ++c) s += vr[c] * S.yb[c];
-    S.tvec[i] = s;
-  }
+PH(4);
+    // ---- E: alpha = V^T (V y), y.alpha ------------------------------
+  for (int i = tid; i < k; i += WG)
+    S.tvec[i] = dot4(S.A + (size_t)i * SA, 1, S.yb, 1, 0, i + 1);
   __syncthreads();
-  for (int a = tid; a < k; a += WG) {
-    float s = 0.f;
-    for (int i = a; i < k; ++i) s += S.A[(size_t)i * SA + a] * S.tvec[i];
-    S.alpha[a] = s;
-  }
+  for (int a = tid; a < k; a += WG)
+    S.alpha[a] = dot4(S.A + a, SA, S.tvec, 1, a, k);
   __syncthreads();
   double part = 0.0;
   for (int i = tid; i < k; i += WG)
     part += (double)S.yb[i] * (double)S.alpha[i];
   const double yta = block_sum(part, S.red, tid);
 
-  // ---- L: K^-1 = V^T V in place (lauum), ascending row blocks ------
-  // element (i, j): sum_{c >= max(i,j)} V[c][i] V[c][j] — valid for any j.
+PH(5);
+    // ---- L: K^-1 = V^T V in place (lauum), ascending row blocks ------
+  // strictly j <= i so the Kb cache in the upper triangle survives
   for (int I = 0; I < nblk; ++I) {
     const int ib = I * NB;
     const int bs = min(NB, k - ib);
-    const int ncol = min(k, ib + bs);        // columns 0 .. ib+bs-1
+    const int ncol = ib + bs;
     for (int f = tid; f < bs * ncol; f += WG) {
       const int r = f / ncol, j = f - r * ncol;
       const int i = ib + r;
-      const int c0 = max(i, j);
-      float s = 0.f;
-      for (int c = c0; c < k; ++c)
-        s += S.A[(size_t)c * SA + i] * S.A[(size_t)c * SA + j];
-      S.T[r * (k + 1) + j] = s;
+      if (j > i) continue;
+      S.T[r * (k + 1) + j] = dot4(S.A + i, SA, S.A + j, SA, i, k);
     }
     __syncthreads();
     for (int f = tid; f < bs * ncol; f += WG) {
       const int r = f / ncol, j = f - r * ncol;
+      if (j > ib + r) continue;
       S.A[(size_t)(ib + r) * SA + j] = S.T[r * (k + 1) + j];
     }
     __syncthreads();
   }
 
-  // ---- W: W0 = (aa^T - K^-1) o Kb, in place + mirror; trG, sumW0 ---
+PH(6);
+    // ---- W: W0 = (aa^T - K^-1) o Kb, in place + mirror; trG, sumW0 ---
   double trg_part = 0.0, sw_part = 0.0;
   {
     const int nlow = k * (k + 1) / 2;
@@ -348,14 +397,8 @@ fused_expert_nll_kernel(const float* __restrict__ Xg,
       int a, b;
       tri_decode(f, a, b);
       const float g = S.alpha[a] * S.alpha[b] - S.A[(size_t)a * SA + b];
-      float q = 0.f;
-      const float* xa = S.X + a * dp;
-      const float* xb = S.X + b * dp;
-      for (int j = 0; j < d; ++j) {
-        float t = xa[j] - xb[j];
-        q += S.s2[j] * t * t;
-      }
-      const float w = g * __expf(-q);
+      const float kb = (a == b) ? 1.0f : S.A[(size_t)b * SA + a];
+      const float w = g * kb;
       if (a == b) {
         trg_part += (double)g;
         sw_part += (double)w;
@@ -363,30 +406,34 @@ fused_expert_nll_kernel(const float* __restrict__ Xg,
         sw_part += 2.0 * (double)w;
       }
       S.A[(size_t)a * SA + b] = w;
-      S.A[(size_t)b * SA + a] = w;
+      if (a != b) S.A[(size_t)b * SA + a] = w;
     }
   }
   const double trG = block_sum(trg_part, S.red, tid);
   const double sumW0 = block_sum(sw_part, S.red, tid);
 
-  // ---- G: row sums of W0 ------------------------------------------
+PH(7);
+    // ---- G: row sums of W0 ------------------------------------------
   for (int a = tid; a < k; a += WG) {
     const float* wr = S.A + (size_t)a * SA;
-    float s = 0.f;
-    for (int b = 0; b < k; ++b) s += wr[b];
-    S.rrow[a] = s;
+    float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
+    int b = 0;
+    for (; b + 3 < k; b += 4) {
+      s0 += wr[b]; s1 += wr[b + 1]; s2 += wr[b + 2]; s3 += wr[b + 3];
+    }
+    for (; b < k; ++b) s0 += wr[b];
+    S.rrow[a] = (s0 + s1) + (s2 + s3);
   }
   __syncthreads();
 
-  // ---- H: WX = W0 @ X (d-halves through T); contraction ------------
+PH(8);
+    // ---- H: WX = W0 @ X (d-halves through T); contraction ------------
   for (int d0 = 0; d0 < d; d0 += 32) {
     const int dl = min(32, d - d0);
     for (int f = tid; f < k * dl; f += WG) {
       const int a = f / dl, j = f - a * dl;
-      const float* wr = S.A + (size_t)a * SA;
-      float s = 0.f;
-      for (int b = 0; b < k; ++b) s += wr[b] * S.X[b * dp + d0 + j];
-      S.T[a * 33 + j] = s;
+      S.T[a * 33 + j] = dot4(S.A + (size_t)a * SA, 1,
+                             S.X + d0 + j, dp, 0, k);
     }
     __syncthreads();
     // contr_j = 2 sum_a x_aj^2 r_a - 2 sum_a x_aj WX_aj   (fp64)
@@ -402,6 +449,7 @@ fused_expert_nll_kernel(const float* __restrict__ Xg,
     __syncthreads();
   }
 
+  PH(9);
   if (tid == 0) {
     out_bad[e] = 0;
     out_nll[e] = 0.5 * yta + 0.5 * logdet;
@@ -415,13 +463,15 @@ extern "C" hipError_t launch_fused_expert_nll(
     const float* X, const float* y, const float* scale,
     float amp, float noise, int E, int k, int d,
     double* out_nll, double* out_sumW0, double* out_trG, double* out_contr,
-    int* out_bad, hipStream_t stream, size_t* lds_used) {
+    int* out_bad, unsigned long long* out_clk, hipStream_t stream,
+    size_t* lds_used) {
   size_t lds = nll_lds_bytes2(k, d);
   if (lds_used) *lds_used = lds;
   if (lds > 160 * 1024 || k > 128 || d > k || d > 64)
     return hipErrorInvalidConfiguration;
   hipLaunchKernelGGL(fused_expert_nll_kernel, dim3(E), dim3(WG), lds, stream,
                      X, y, scale, amp, noise, k, d,
-                     out_nll, out_sumW0, out_trG, out_contr, out_bad);
+                     out_nll, out_sumW0, out_trG, out_contr, out_bad,
+                     out_clk);
   return hipGetLastError();
 }
