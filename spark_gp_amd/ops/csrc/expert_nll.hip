@@ -34,7 +34,7 @@
 // whose fp32 Cholesky breaks down are flagged in out_bad and recomputed on
 // the torch fallback path by the host.
 //
-// Constraints: k <= 128, d <= min(k, 64); LDS budget checked host-side.
+// Constraints: k <= 128, d <= 128, LDS budget checked host-side.
 
 #include <hip/hip_runtime.h>
 #include <math.h>
@@ -467,7 +467,7 @@ extern "C" hipError_t launch_fused_expert_nll(
     size_t* lds_used) {
   size_t lds = nll_lds_bytes2(k, d);
   if (lds_used) *lds_used = lds;
-  if (lds > 160 * 1024 || k > 128 || d > k || d > 64)
+  if (lds > 160 * 1024 || k > 128 || d > 128)
     return hipErrorInvalidConfiguration;
   hipLaunchKernelGGL(fused_expert_nll_kernel, dim3(E), dim3(WG), lds, stream,
                      X, y, scale, amp, noise, k, d,

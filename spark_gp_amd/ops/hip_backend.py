@@ -49,12 +49,13 @@ def nll_grad_compiled(cs: CompiledKernel, theta: np.ndarray,
     nll, sumW0, trG, contr, bad = ext.fused_expert_nll(
         X, y.to(torch.float32), scale, float(C), float(nu))
 
-    bad_mask = bad != 0
-    n_bad = int(bad_mask.sum())
-    nll_total = float(nll.sum())
-    sumW0_t = float(sumW0.sum())
-    trG_t = float(trG.sum())
-    contr_t = contr.sum(0).cpu().numpy()              # [d]
+    # single device->host transfer for all reductions (one sync per eval)
+    stats = torch.cat([nll.sum().reshape(1), sumW0.sum().reshape(1),
+                       trG.sum().reshape(1), bad.sum().double().reshape(1),
+                       contr.sum(0)]).cpu().numpy()
+    nll_total, sumW0_t, trG_t = float(stats[0]), float(stats[1]), float(stats[2])
+    n_bad = int(stats[3])
+    contr_t = stats[4:]                               # [d]
 
     grad = np.zeros(cs.p)
     if cs.amp_idx is not None:
@@ -72,7 +73,7 @@ def nll_grad_compiled(cs: CompiledKernel, theta: np.ndarray,
     if n_bad:
         # fp32 Cholesky broke down for these experts (huge-amplitude
         # iterates); recompute them on the torch path (LU fallback inside)
-        idx = bad_mask.nonzero(as_tuple=True)[0]
+        idx = (bad != 0).nonzero(as_tuple=True)[0]
         nll_b, grad_b = torch_backend.nll_grad_compiled(
             cs, theta, X[idx], y[idx])
         nll_total += nll_b
