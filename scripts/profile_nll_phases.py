@@ -15,8 +15,9 @@ X = torch.rand(E, k, d, generator=g).cuda()
 y = torch.sin(3 * X.sum(-1)).cuda()
 scale = torch.rand(d, generator=g).add(0.5).cuda()
 
-# warmup + timing
-for _ in range(2):
+# warmup + DVFS clock ramp (sustained load for ~10s)
+t0 = time.perf_counter()
+while time.perf_counter() - t0 < 10.0:
     out = ext.fused_expert_nll(X, y, scale, 1.0, 1e-3)
 torch.cuda.synchronize()
 t0 = time.perf_counter()

@@ -10,11 +10,15 @@
 //
 //   B  lower(A) = amp*Kb + noise*I,  strict upper(A) = Kb (cached base
 //      kernel — read back in phase W, never recomputed)
-//   C  blocked right-looking Cholesky, IN PLACE: per 32-column block, the
-//      diagonal block is factored AND inverted by one wave (lockstep
-//      __syncwarp steps, both 32-lane halves cooperating), the panel solve
-//      becomes a dense GEMM against the inverted diagonal, the trailing
-//      update is an all-thread data-parallel GEMM.  fp32; fp64 logdet.
+//   C  blocked right-looking Cholesky, IN PLACE, with look-ahead: per
+//      32-column block, wave 0 factors AND inverts the diagonal block
+//      (8x8 sub-diagonals factored+inverted entirely inside lane 0's
+//      registers — the serial dependency chain never touches LDS — with
+//      lane-parallel 8-wide panels/trailing and block back-substitution
+//      assembling the 32x32 inverse) WHILE waves 1-7 apply the previous
+//      panel's rank-32 trailing update to the remaining columns; the
+//      panel solve is a dense GEMM against the inverted diagonal.
+//      fp32; fp64 logdet.
 //   D  blocked in-place triangular inverse of the off-diagonal blocks
 //      (right-to-left column blocks, rows fully parallel:
 //       V_IJ = -(sum_{K>J} V_IK L_KJ) L_JJ^-1).
@@ -198,89 +202,199 @@ PH(1);
   __syncthreads();
 
 PH(2);
-    // ---- C: blocked in-place Cholesky + diagonal-block inverse --------
+    // ---- C: blocked in-place Cholesky with look-ahead ----------------
+  // Right-looking, restructured so the serial diagonal factorization
+  // overlaps the data-parallel trailing update: per column block J,
+  //   phase1: apply panel J-1's rank-NB update to COLUMN-BLOCK J only
+  //   phase2: wave 0 factors+inverts diag J  ||  waves 1-7 apply panel
+  //           J-1's update to the remaining trailing columns
+  //   phase3: panel solve for block J (copy + GEMM vs inverted diagonal)
   for (int J = 0; J < nblk; ++J) {
     const int jb = J * NB;
     const int bs = min(NB, k - jb);
     float* D = S.A + (size_t)jb * SA + jb;   // diag block, stride SA
+    const int pj = jb - NB;                  // previous panel column offset
+
+    if (J > 0) {
+      // phase1: update column-block J (rows jb..k, cols jb..jb+bs, c<=i)
+      const int rows = k - jb;
+      for (int f = tid; f < rows * bs; f += WG) {
+        const int r = f / bs, c = f - r * bs;
+        const int i = jb + r, cc = jb + c;
+        if (cc > i) continue;                // keep the Kb upper cache
+        S.A[(size_t)i * SA + cc] -=
+            dot4(S.A + (size_t)i * SA + pj, 1,
+                 S.A + (size_t)cc * SA + pj, 1, 0, NB);
+      }
+      __syncthreads();
+    }
 
     unsigned long long c1t0 = 0;
     if (out_clk && tid == 0) c1t0 = wall_clock64();
     if (tid < 64) {
-      // wave 0: register-resident potrf + trtri of the bs x bs diagonal
-      // block, fully unrolled so every register-array index is static and
-      // the whole factorization runs on VGPRs + cross-lane shfl — no LDS
-      // round-trips in the serial dependency chain.
-      const int row = lane & 31;             // both 32-lane halves mirror
-      const int half = lane >> 5;
-      {
-        // potrf with deferred column scaling: the trailing update works on
-        // RAW columns (A[i][c] -= A[i][s] A[c][s] / a_ss), so each step's
-        // only cross-step LDS dependency is the pivot read -> ~2 LDS hops
-        // per step instead of 4.  Columns are scaled to L in one parallel
-        // pass at the end (pivots survive: column s is final at step s).
-        double ldet = 0.0;
-        bool ok = true;
-        for (int ss = 0; ss < bs; ++ss) {
-          const float ajj = D[ss * SA + ss];   // broadcast LDS read
-          if (!(ajj > 0.f) || !isfinite(ajj)) { ok = false; break; }
-          ldet += (double)__logf(ajj);
-          const float inv = __builtin_amdgcn_rcpf(ajj);
-          if (row > ss && row < bs) {
-            const float ris = D[row * SA + ss] * inv;
-            // trailing rank-1 update; the two 32-lane halves split by
-            // parity; 4-wide batches so loads share one lgkm wait
-            float* wr = D + (size_t)row * SA;
-            const float* pc = D + ss;
-            int c = ss + 1 + half;
-            for (; c + 6 <= row; c += 8) {
-              const float p0 = pc[c * SA], p1 = pc[(c + 2) * SA];
-              const float p2 = pc[(c + 4) * SA], p3 = pc[(c + 6) * SA];
-              const float a0 = wr[c], a1 = wr[c + 2];
-              const float a2 = wr[c + 4], a3 = wr[c + 6];
-              wr[c] = a0 - ris * p0;
-              wr[c + 2] = a1 - ris * p1;
-              wr[c + 4] = a2 - ris * p2;
-              wr[c + 6] = a3 - ris * p3;
-            }
-            for (; c <= row; c += 2)
-              wr[c] -= ris * pc[c * SA];
-          }
-          __syncwarp();
-        }
-        if (ok) {
-          // scale pass: L[i][s] = raw[i][s] * rsqrt(raw[s][s]) (diag incl.)
-          if (lane < bs) S.T[lane] = rsqrtf(D[lane * SA + lane]);
-          __syncwarp();
-          const int ntri = bs * (bs + 1) / 2;
-          for (int f = lane; f < ntri; f += 64) {
-            int i, c;
-            tri_decode(f, i, c);
-            D[i * SA + c] *= S.T[c];
-          }
-          __syncwarp();
-        }
+      // wave 0: factor + invert the bs x bs diagonal block via 8x8
+      // sub-blocks.  Each 8x8 sub-diagonal is Cholesky-factored AND
+      // inverted entirely inside lane 0's registers (fully unrolled, the
+      // serial dependency chain never touches LDS); panels and trailing
+      // updates are lane-parallel; the 32-level inverse assembles from the
+      // 8x8 inverses by block back-substitution.  8x8 inverses live in
+      // T[0..256); T[256..448) is shell scratch.
+      const int row = lane & 31;
+      const int nq = (bs + 7) / 8;
+      float* Vq = S.T;               // [nq][8][8]
+      float* TS = S.T + 256;         // shell scratch [3][8][8]
+      for (int q = 0; q < nq; ++q) {
+        const int qb = q * 8;
+        const int sbs = min(8, bs - qb);
         if (lane == 0) {
-          if (ok) S.misc[0] += ldet;
-          else *S.bad = 1;
+          float m[8][8];
+#pragma unroll
+          for (int i = 0; i < 8; ++i)
+#pragma unroll
+            for (int c = 0; c < 8; ++c)
+              m[i][c] = (i < sbs && c <= i)
+                            ? D[(size_t)(qb + i) * SA + qb + c]
+                            : (i == c ? 1.f : 0.f);
+          double ldet = 0.0;
+          bool ok = true;
+#pragma unroll
+          for (int ss = 0; ss < 8; ++ss) {
+            const float piv = m[ss][ss];
+            if (ss < sbs) {
+              if (!(piv > 0.f) || !isfinite(piv)) ok = false;
+              else ldet += (double)__logf(piv);
+            }
+            const float rs = rsqrtf(piv);
+            m[ss][ss] = piv * rs;
+#pragma unroll
+            for (int i = ss + 1; i < 8; ++i) m[i][ss] *= rs;
+#pragma unroll
+            for (int i = ss + 1; i < 8; ++i)
+#pragma unroll
+              for (int c = ss + 1; c <= i; ++c)
+                m[i][c] -= m[i][ss] * m[c][ss];
+          }
+          // write L back; in-register trtri8 into Vq
+#pragma unroll
+          for (int i = 0; i < 8; ++i)
+#pragma unroll
+            for (int c = 0; c <= i; ++c)
+              if (i < sbs) D[(size_t)(qb + i) * SA + qb + c] = m[i][c];
+          float v8[8][8];
+#pragma unroll
+          for (int i = 0; i < 8; ++i) {
+            const float rli = __builtin_amdgcn_rcpf(m[i][i]);
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+              if (j > i) { v8[i][j] = 0.f; continue; }
+              if (j == i) { v8[i][j] = rli; continue; }
+              float sacc = 0.f;
+#pragma unroll
+              for (int c = 0; c < 8; ++c)
+                if (c >= j && c < i) sacc += m[i][c] * v8[c][j];
+              v8[i][j] = -sacc * rli;
+            }
+          }
+#pragma unroll
+          for (int i = 0; i < 8; ++i)
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+              Vq[q * 64 + i * 8 + j] = v8[i][j];
+          if (!ok) *S.bad = 1;
+          else S.misc[0] += ldet;
+        }
+        __syncwarp();
+        // panel rows within the 32-block: P = A * Vq^T; then trailing
+        const int p0 = qb + sbs;       // first panel row (local)
+        const int pr = bs - p0;        // panel rows
+        if (pr > 0) {
+          // panel: (pr x sbs) elements, lanes parallel
+          float pv[1];
+          (void)pv;
+          for (int f = lane; f < pr * 8; f += 64) {
+            const int r = f >> 3, c = f & 7;
+            if (c >= sbs) continue;
+            float sacc = 0.f;
+            const float* ar = D + (size_t)(p0 + r) * SA + qb;
+            const float* vr = Vq + q * 64 + c * 8;
+            for (int t = 0; t <= c; ++t) sacc += ar[t] * vr[t];
+            TS[f] = sacc;              // hold row piece until all reads done
+          }
+          __syncwarp();
+          for (int f = lane; f < pr * 8; f += 64) {
+            const int r = f >> 3, c = f & 7;
+            if (c >= sbs) continue;
+            D[(size_t)(p0 + r) * SA + qb + c] = TS[f];
+          }
+          __syncwarp();
+          // trailing: lower incl diag of remaining rows
+          const int ntri = pr * (pr + 1) / 2;
+          for (int f = lane; f < ntri; f += 64) {
+            int a, b;
+            tri_decode(f, a, b);
+            const int i = p0 + a, c = p0 + b;
+            S.A[(size_t)(jb + i) * SA + jb + c] -=
+                dot4(D + (size_t)i * SA + qb, 1, D + (size_t)c * SA + qb, 1,
+                     0, sbs);
+          }
+          __syncwarp();
         }
       }
-      __syncwarp();
-      if (!*S.bad) {
-        // trtri in place: step i reads L row i (broadcast) and the V
-        // entries of earlier steps (already written into D), then
-        // overwrites row i with V row i.  Wave lockstep orders the
-        // within-step reads before the write.
-        for (int i = 0; i < bs; ++i) {
-          const float* Li = D + i * SA;
-          const float sacc = dot4(Li, 1, D + row, SA, row, i);
-          const float rli = __builtin_amdgcn_rcpf(Li[i]);
-          const float vi = (row < i) ? (-sacc * rli)
-                                     : (row == i ? rli : 0.f);
-          __syncwarp();
-          if (lane < 32 && row <= i) D[i * SA + row] = vi;
-          __syncwarp();
+      // ---- assemble V_JJ (32x32 inverse) from the 8x8 inverses --------
+      // column-blocks Jq descending; all target blocks of a column in
+      // parallel through TS
+      for (int Jq = nq - 1; Jq >= 0; --Jq) {
+        const int jb8 = Jq * 8;
+        const int nblks = nq - 1 - Jq;       // target blocks below
+        for (int f = lane; f < nblks * 64; f += 64) {
+          const int blk = f >> 6;            // 0..nblks-1
+          const int ib = (Jq + 1 + blk) * 8;
+          const int i = (f >> 3) & 7, j = f & 7;
+          const int gi = ib + i;
+          float u = 0.f;
+          if (gi < bs && jb8 + j < bs) {
+            // u = sum_{c=jb8+8}^{gi} V[gi][c] * L[c][jb8+j]
+            u = dot4(D + (size_t)gi * SA, 1, D + jb8 + j, SA,
+                     jb8 + 8, min(gi + 1, bs));
+          }
+          TS[f] = u;
         }
+        __syncwarp();
+        for (int f = lane; f < nblks * 64; f += 64) {
+          const int blk = f >> 6;
+          const int ib = (Jq + 1 + blk) * 8;
+          const int i = (f >> 3) & 7, j = f & 7;
+          const int gi = ib + i;
+          if (gi >= bs || jb8 + j >= bs) continue;
+          float sacc = 0.f;
+          const float* ts = TS + blk * 64 + i * 8;
+          const float* vv = Vq + Jq * 64;
+#pragma unroll
+          for (int t = 0; t < 8; ++t) sacc += ts[t] * vv[t * 8 + j];
+          D[(size_t)gi * SA + jb8 + j] = -sacc;
+        }
+        // this column's diagonal block <- its inverse (consumed as V by
+        // the shells of columns further left)
+        for (int f = lane; f < 64; f += 64) {
+          const int i = (f >> 3) & 7, j = f & 7;
+          const int gi = jb8 + i, gj = jb8 + j;
+          if (gi < bs && gj <= gi) D[(size_t)gi * SA + gj] = Vq[Jq * 64 + f];
+        }
+        __syncwarp();
+      }
+      __syncwarp();
+    } else if (J > 0) {    } else if (J > 0) {
+      // waves 1-7: previous panel's trailing update beyond column-block J
+      const int t0r = jb + bs;
+      const int tr = k - t0r;
+      const int ntri = tr * (tr + 1) / 2;
+      for (int f = tid - 64; f < ntri; f += WG - 64) {
+        int a, b;
+        tri_decode(f, a, b);
+        const int i = t0r + a, c = t0r + b;
+        S.A[(size_t)i * SA + c] -=
+            dot4(S.A + (size_t)i * SA + pj, 1,
+                 S.A + (size_t)c * SA + pj, 1, 0, NB);
       }
     }
     if (out_clk && tid == 0)
@@ -302,17 +416,6 @@ PH(2);
         int r = f / bs, c = f - r * bs;
         S.A[(size_t)(t0 + r) * SA + jb + c] =
             dot4(S.T + r * 33, 1, D + c * SA, 1, 0, c + 1);
-      }
-      __syncthreads();
-      // C3: trailing update (lower incl. diag): A[i][c] -= L[i][Jb].L[c][Jb]
-      const int ntri = nr * (nr + 1) / 2;
-      for (int f = tid; f < ntri; f += WG) {
-        int a, b;
-        tri_decode(f, a, b);
-        const int i = t0 + a, c = t0 + b;
-        S.A[(size_t)i * SA + c] -=
-            dot4(S.A + (size_t)i * SA + jb, 1, S.A + (size_t)c * SA + jb, 1,
-                 0, bs);
       }
       __syncthreads();
     }
