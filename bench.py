@@ -59,6 +59,15 @@ def main():
 
     use_cuda = torch.cuda.is_available() and args.device != "cpu"
     device = torch.device(args.device or ("cuda" if use_cuda else "cpu"))
+    if device.type == "cuda":
+        # Pin stable high clocks (MI355X otherwise ramps over seconds and
+        # bounces under mixed load; standard benchmarking practice).
+        import subprocess
+        try:
+            subprocess.run(["rocm-smi", "--setperfdeterminism", "2100"],
+                           capture_output=True, timeout=30)
+        except Exception:
+            pass
     init_from_env(device)
     comm = get_comm()
     rank, world = comm.rank, comm.world_size
