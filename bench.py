@@ -45,6 +45,10 @@ def parse_args():
     p.add_argument("--ppa-precision", type=str, default="mixed",
                    choices=["mixed", "fp64"],
                    help="PPA SYRK path: hi/lo bf16 MFMA (mixed) or fp64")
+    p.add_argument("--optimizer-restart", action="store_true",
+                   help="enable the restart-on-bound-collapse guard inside "
+                        "each fit (off by default: the reference runs one "
+                        "L-BFGS-B solve)")
     p.add_argument("--min-warmup-seconds", type=float, default=8.0,
                    help="keep running warmup fits until this much wall time "
                         "has passed (DVFS clock stabilization)")
@@ -91,6 +95,7 @@ def main():
                 .setMaxIter(args.max_iter)
                 .setSeed(args.seed)
                 .setPpaPrecision(args.ppa_precision)
+                .setOptimizerRestart(args.optimizer_restart)
                 .setDevice(str(device)))
 
     def sync():
@@ -138,7 +143,8 @@ def main():
             "data": "synthetic",
             "config": {
                 "model": f"ARD-RBF GP regression, m={args.active_set}, "
-                         f"expert={args.expert_size}, maxIter={args.max_iter}",
+                         f"expert={args.expert_size}, maxIter={args.max_iter}, "
+                         f"restart={args.optimizer_restart}",
                 "rows": args.rows,
                 "dim": args.dim,
                 "active_set": args.active_set,

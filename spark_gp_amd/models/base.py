@@ -79,6 +79,7 @@ class GaussianProcessParams:
         self._dtype: Optional[torch.dtype] = None  # None -> f64 CPU / f32 GPU
         self._ppa_precision = "fp64"  # 'fp64' (reference-parity) | 'mixed'
                                       # (hi/lo bf16 MFMA SYRK, fastest)
+        self._optimizer_restart = True  # restart-on-bound-collapse guard
 
     # Reference-parity camelCase setters -------------------------------
     def setKernel(self, factory: Callable[[], Kernel]):
@@ -116,6 +117,12 @@ class GaussianProcessParams:
     # Additive (no reference analog)
     def setDevice(self, device: str):
         self._device = device
+        return self
+
+    def setOptimizerRestart(self, v: bool):
+        """Enable/disable the restart-on-bound-collapse guard (an additive
+        robustness feature; the reference runs exactly one L-BFGS-B solve)."""
+        self._optimizer_restart = bool(v)
         return self
 
     def setPpaPrecision(self, p: str):
@@ -190,7 +197,8 @@ class GaussianProcessCommons(GaussianProcessParams):
 
         t0 = time.perf_counter()
         opt = lbfgsb(objective, x0, lower, upper,
-                     max_iter=self._max_iter, tol=self._tol)
+                     max_iter=self._max_iter, tol=self._tol,
+                     restart_on_bound_collapse=self._optimizer_restart)
         instr.log_timing("optimize_hypers", time.perf_counter() - t0)
         optimal = self._get_kernel().set_hyperparameters(opt)
         instr.log(f"Optimal kernel: {optimal!r}")

@@ -216,12 +216,13 @@ PH(2);
     const int pj = jb - NB;                  // previous panel column offset
 
     if (J > 0) {
-      // phase1: update column-block J (rows jb..k, cols jb..jb+bs, c<=i)
-      const int rows = k - jb;
-      for (int f = tid; f < rows * bs; f += WG) {
+      // phase1: update ONLY the diagonal block (rows/cols jb..jb+bs, c<=i)
+      // so wave 0 can start factoring immediately; the rest of column-block
+      // J and the trailing matrix are updated by waves 1-7 during phase2.
+      for (int f = tid; f < bs * bs; f += WG) {
         const int r = f / bs, c = f - r * bs;
+        if (c > r) continue;                 // keep the Kb upper cache
         const int i = jb + r, cc = jb + c;
-        if (cc > i) continue;                // keep the Kb upper cache
         S.A[(size_t)i * SA + cc] -=
             dot4(S.A + (size_t)i * SA + pj, 1,
                  S.A + (size_t)cc * SA + pj, 1, 0, NB);
@@ -384,10 +385,18 @@ PH(2);
       }
       __syncwarp();
     } else if (J > 0) {    } else if (J > 0) {
-      // waves 1-7: previous panel's trailing update beyond column-block J
+      // waves 1-7: previous panel's update to (a) the panel rows of
+      // column-block J and (b) the remaining trailing triangle
       const int t0r = jb + bs;
-      const int tr = k - t0r;
-      const int ntri = tr * (tr + 1) / 2;
+      const int nrp = k - t0r;
+      for (int f = tid - 64; f < nrp * bs; f += WG - 64) {
+        const int r = f / bs, c = f - r * bs;
+        const int i = t0r + r, cc = jb + c;
+        S.A[(size_t)i * SA + cc] -=
+            dot4(S.A + (size_t)i * SA + pj, 1,
+                 S.A + (size_t)cc * SA + pj, 1, 0, NB);
+      }
+      const int ntri = nrp * (nrp + 1) / 2;
       for (int f = tid - 64; f < ntri; f += WG - 64) {
         int a, b;
         tri_decode(f, a, b);
