@@ -118,8 +118,12 @@ def main():
 
     sync()
     t0 = time.perf_counter()
-    for _ in range(args.steps):
+    for si in range(args.steps):
         model = make_gp().fit(Xt, yt)
+        if os.environ.get("SPARK_GP_BENCH_VERBOSE") == "1" and rank == 0:
+            print(f"step {si}:", {k: round(v, 4)
+                                  for k, v in model._instr.timings.items()},
+                  flush=True)
     sync()
     elapsed = time.perf_counter() - t0
 

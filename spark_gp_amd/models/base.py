@@ -189,10 +189,15 @@ class GaussianProcessCommons(GaussianProcessParams):
         x0 = kernel.get_hyperparameters()
         lower, upper = kernel.hyperparameter_bounds()
 
+        evals = [0, 0.0]
+
         def objective(theta: np.ndarray) -> Tuple[float, np.ndarray]:
+            te = time.perf_counter()
             nll, grad = local_obj(theta)
             buf = np.concatenate([[nll], grad])
             buf = comm.allreduce_np(buf)
+            evals[0] += 1
+            evals[1] += time.perf_counter() - te
             return float(buf[0]), buf[1:]
 
         t0 = time.perf_counter()
@@ -200,6 +205,8 @@ class GaussianProcessCommons(GaussianProcessParams):
                      max_iter=self._max_iter, tol=self._tol,
                      restart_on_bound_collapse=self._optimizer_restart)
         instr.log_timing("optimize_hypers", time.perf_counter() - t0)
+        instr.timings["objective_evals"] = evals[0]
+        instr.timings["objective_time"] = evals[1]
         optimal = self._get_kernel().set_hyperparameters(opt)
         instr.log(f"Optimal kernel: {optimal!r}")
         return opt

@@ -119,7 +119,8 @@ cross_kernel_tile_kernel(const float* __restrict__ X,    // [c, d]
 // permutation).  Fragment loads are single ds_read_b128 each.
 
 #define SY_BK 32
-#define SY_STR 40     // LDS row stride in bf16 elements (80 B)
+#define SY_STR 136    // row-major [SY_BK][SY_STR] bf16; 272-B rows (16-B
+                      // aligned so staging is plain b128 writes)
 
 template <bool HILO>
 __global__ void __launch_bounds__(256)
@@ -129,17 +130,19 @@ syrk_bf16_kernel(const bf16* __restrict__ Kc,   // [c, m] hi part
                  const int split_k,
                  float* __restrict__ KK) {      // [m, m] accumulated
   // with HILO: KK += hi^T hi + hi^T lo + lo^T hi  (lo^T lo ~ 2^-32, dropped)
-  __shared__ __align__(16) bf16 lt[128 * SY_STR];
-  __shared__ __align__(16) bf16 rt[128 * SY_STR];
-  __shared__ __align__(16) bf16 ltl[HILO ? 128 * SY_STR : 1];
-  __shared__ __align__(16) bf16 rtl[HILO ? 128 * SY_STR : 1];
+  // Tiles stage ROW-major ([k][col]): global loads are coalesced 16-B
+  // vectors and LDS writes are conflict-free b128; the k-strided MFMA
+  // fragment gathers are scalar b16 reads (2-way worst case).
+  __shared__ __align__(16) bf16 lt[SY_BK * SY_STR];
+  __shared__ __align__(16) bf16 rt[SY_BK * SY_STR];
+  __shared__ __align__(16) bf16 ltl[HILO ? SY_BK * SY_STR : 1];
+  __shared__ __align__(16) bf16 rtl[HILO ? SY_BK * SY_STR : 1];
 
   const int tile = blockIdx.x;
   const int ti = tile / ntile, tj = tile % ntile;
   if (tj < ti) return;                     // upper-triangle tiles only
   const int i0 = ti * 128, j0 = tj * 128;
   const int slice = blockIdx.y;
-  // K-range of this split-K slice (multiples of SY_BK)
   const int kblocks = (c + SY_BK - 1) / SY_BK;
   const int per = (kblocks + split_k - 1) / split_k;
   const int kb0 = slice * per;
@@ -160,20 +163,24 @@ syrk_bf16_kernel(const bf16* __restrict__ Kc,   // [c, m] hi part
 
   const int l16 = lane & 15;              // fragment row/col
   const int kgrp = lane >> 4;             // 0..3 -> k-subblock of 8
+  const bool mvec = (m % 8 == 0);         // 16-B global loads legal
 
+  // staging assignment: thread -> (k-row, 8-col segment); 512 slots per
+  // operand pair = 2 iterations of 256 threads
   for (int kb = kb0; kb < kb1; ++kb) {
     const int krow0 = kb * SY_BK;
-    // stage both tiles k-major: lt[i][kk] = Kc[krow0+kk][i0+i]
-    // 256 threads x 16 elems: each thread loads 8 consecutive columns of
-    // one k-row for each tile (vector global load, scattered LDS writes)
-    for (int f = tid; f < SY_BK * 16; f += 256) {
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      const int f = tid + half * 256;
       const int kk = f >> 4;              // 0..31
-      const int seg = f & 15;             // 16 segments of 8 columns
+      const int seg = f & 15;
       const int col = seg * 8;
       const int gk = krow0 + kk;
-      bf16 vals[8];
       auto stage = [&](const bf16* mat, int c0, bf16* dst) {
-        if (gk < c) {
+        bf16 vals[8];
+        if (gk < c && mvec && c0 + col + 8 <= m) {
+          *(uint4*)vals = *(const uint4*)(mat + (size_t)gk * m + c0 + col);
+        } else if (gk < c) {
           const bf16* src = mat + (size_t)gk * m;
 #pragma unroll
           for (int u = 0; u < 8; ++u) {
@@ -184,8 +191,7 @@ syrk_bf16_kernel(const bf16* __restrict__ Kc,   // [c, m] hi part
 #pragma unroll
           for (int u = 0; u < 8; ++u) vals[u] = (bf16)0.f;
         }
-#pragma unroll
-        for (int u = 0; u < 8; ++u) dst[(col + u) * SY_STR + kk] = vals[u];
+        *(uint4*)&dst[kk * SY_STR + col] = *(uint4*)vals;
       };
       stage(Kc, i0, lt);
       stage(Kc, j0, rt);
@@ -196,23 +202,33 @@ syrk_bf16_kernel(const bf16* __restrict__ Kc,   // [c, m] hi part
     }
     __syncthreads();
 
-    // 2 k-steps of 32 wait: mfma 16x16x32 consumes K=32 per instruction;
-    // our BK=32 is exactly one instruction depth per (a, b) pair.
     {
 #pragma unroll
       for (int a = 0; a < 4; ++a) {
         const int arow = wr + a * 16 + l16;
-        bf16x8 afrag = *(const bf16x8*)&lt[arow * SY_STR + kgrp * 8];
-        bf16x8 afl;
-        if (HILO) afl = *(const bf16x8*)&ltl[arow * SY_STR + kgrp * 8];
+        bf16x8 afrag, afl;
+#pragma unroll
+        for (int t = 0; t < 8; ++t)
+          afrag[t] = lt[(kgrp * 8 + t) * SY_STR + arow];
+        if (HILO) {
+#pragma unroll
+          for (int t = 0; t < 8; ++t)
+            afl[t] = ltl[(kgrp * 8 + t) * SY_STR + arow];
+        }
 #pragma unroll
         for (int b = 0; b < 4; ++b) {
           const int bcol = wc + b * 16 + l16;
-          bf16x8 bfrag = *(const bf16x8*)&rt[bcol * SY_STR + kgrp * 8];
+          bf16x8 bfrag;
+#pragma unroll
+          for (int t = 0; t < 8; ++t)
+            bfrag[t] = rt[(kgrp * 8 + t) * SY_STR + bcol];
           acc[a][b] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afrag, bfrag, acc[a][b], 0, 0, 0);
           if (HILO) {
-            bf16x8 bfl = *(const bf16x8*)&rtl[bcol * SY_STR + kgrp * 8];
+            bf16x8 bfl;
+#pragma unroll
+            for (int t = 0; t < 8; ++t)
+              bfl[t] = rtl[(kgrp * 8 + t) * SY_STR + bcol];
             acc[a][b] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                 afrag, bfl, acc[a][b], 0, 0, 0);
             acc[a][b] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(

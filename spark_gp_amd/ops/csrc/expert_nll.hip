@@ -248,6 +248,7 @@ PH(2);
         const int qb = q * 8;
         const int sbs = min(8, bs - qb);
         if (lane == 0) {
+          bool nonfin = false;
           float m[8][8];
 #pragma unroll
           for (int i = 0; i < 8; ++i)
@@ -262,7 +263,8 @@ PH(2);
           for (int ss = 0; ss < 8; ++ss) {
             const float piv = m[ss][ss];
             if (ss < sbs) {
-              if (!(piv > 0.f) || !isfinite(piv)) ok = false;
+              if (!isfinite(piv)) { ok = false; nonfin = true; }
+              else if (!(piv > 0.f)) ok = false;
               else ldet += (double)__logf(piv);
             }
             const float rs = rsqrtf(piv);
@@ -301,7 +303,7 @@ PH(2);
 #pragma unroll
             for (int j = 0; j < 8; ++j)
               Vq[q * 64 + i * 8 + j] = v8[i][j];
-          if (!ok) *S.bad = 1;
+          if (!ok) *S.bad = nonfin ? 2 : 1;   // 2: non-finite iterate
           else S.misc[0] += ldet;
         }
         __syncwarp();

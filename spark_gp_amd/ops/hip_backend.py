@@ -52,10 +52,16 @@ def nll_grad_compiled(cs: CompiledKernel, theta: np.ndarray,
     # single device->host transfer for all reductions (one sync per eval)
     stats = torch.cat([nll.sum().reshape(1), sumW0.sum().reshape(1),
                        trG.sum().reshape(1), bad.sum().double().reshape(1),
+                       bad.max().double().reshape(1),
                        contr.sum(0)]).cpu().numpy()
     nll_total, sumW0_t, trG_t = float(stats[0]), float(stats[1]), float(stats[2])
     n_bad = int(stats[3])
-    contr_t = stats[4:]                               # [d]
+    if stats[4] >= 2.0:
+        # a non-finite kernel matrix (overflowing line-search iterate): the
+        # reference's LAPACK would propagate NaN and L-BFGS-B backtracks;
+        # return +inf instead of recomputing every expert on the fallback
+        return float("inf"), np.zeros(cs.p)
+    contr_t = stats[5:]                               # [d]
 
     grad = np.zeros(cs.p)
     if cs.amp_idx is not None:
@@ -75,7 +81,7 @@ def nll_grad_compiled(cs: CompiledKernel, theta: np.ndarray,
         # iterates); recompute them on the torch path (LU fallback inside)
         idx = (bad != 0).nonzero(as_tuple=True)[0]
         nll_b, grad_b = torch_backend.nll_grad_compiled(
-            cs, theta, X[idx], y[idx])
+            cs, theta, X[idx], y[idx], force_lu=True)
         nll_total += nll_b
         grad += grad_b
     return nll_total, grad

@@ -26,13 +26,19 @@ from ..kernels.base import Kernel, sqdist
 from ..kernels.compiled import CompiledKernel
 
 
-def logdet_and_inv(K: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
+def logdet_and_inv(K: torch.Tensor, force_lu: bool = False
+                   ) -> Tuple[torch.Tensor, torch.Tensor]:
     """(logdet [...,], Kinv [..., k, k]) from one factorization per matrix.
 
     Mirrors ``commons/util/logDetAndInv.scala`` (single LU there).  Here:
     Cholesky (cheaper, PD-correct); experts whose Cholesky fails fall back to
     LU-based slogdet+inv so non-PD iterates behave like the reference instead
-    of aborting (the reference's LU never fails on merely-indefinite K)."""
+    of aborting (the reference's LU never fails on merely-indefinite K).
+    ``force_lu`` skips the Cholesky attempt (used when the caller already
+    knows it breaks down, e.g. the HIP kernel's bad-expert fallback)."""
+    if force_lu:
+        _, logabsdet = torch.linalg.slogdet(K)
+        return logabsdet, torch.linalg.inv(K)
     L, info = torch.linalg.cholesky_ex(K)
     bad = info > 0
     if bad.any():
@@ -73,8 +79,8 @@ def _base_matrices(cs: CompiledKernel, theta: np.ndarray, X: torch.Tensor):
 
 
 def nll_grad_compiled(cs: CompiledKernel, theta: np.ndarray,
-                      X: torch.Tensor, y: torch.Tensor
-                      ) -> Tuple[float, np.ndarray]:
+                      X: torch.Tensor, y: torch.Tensor,
+                      force_lu: bool = False) -> Tuple[float, np.ndarray]:
     """Sum over the expert batch of the per-expert BCM negative log marginal
     likelihood and its gradient w.r.t. the full hyperparameter vector.
 
@@ -89,7 +95,7 @@ def nll_grad_compiled(cs: CompiledKernel, theta: np.ndarray,
     k = X.shape[-2]
     K = C * Kb + nu * torch.eye(k, dtype=X.dtype, device=X.device)
 
-    logdet, Kinv = logdet_and_inv(K)
+    logdet, Kinv = logdet_and_inv(K, force_lu=force_lu)
     alpha = (Kinv @ y.unsqueeze(-1)).squeeze(-1)                 # [E, k]
     nll = (0.5 * (y * alpha).sum(-1).double() + 0.5 * logdet.double()).sum()
 
