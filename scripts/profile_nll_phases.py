@@ -30,17 +30,18 @@ print(f"E={E} k={k} d={d}: {dt*1e3:.2f} ms/launch, "
 
 *_, clk = ext.fused_expert_nll_profile(X, y, scale, 1.0, 1e-3, True)
 clk = clk.cpu().numpy().astype(np.float64)
-names = ["stage", "B build", "C chol", "D trtri", "E alpha", "L lauum",
-         "W w0", "G rowsum", "H contr", "out"]
+# deltas[i] = clk[i+1]-clk[i] = duration of phase phases[i]
+phases = ["A stage", "B build", "C chol", "D trtri", "E alpha", "L lauum",
+          "W w0", "G rowsum", "H contr"]
+names = phases
 # per-expert deltas in microseconds (wall_clock64 = 100 MHz)
 FREQ = 1e8
 deltas = (clk[:, 1:10] - clk[:, 0:9]) / FREQ * 1e6
 mean = deltas.mean(0)
 total = mean.sum()
 print(f"sum of phases: {total:.2f} us (mean per expert)")
-for i, n in enumerate(names[:9]):
-    print(f"  {names[i+1] if False else n:>8} -> {names[i+1]:>8}: "
-          f"{mean[i]:8.2f} us  ({100*mean[i]/total:4.1f}%)")
+for i in range(9):
+    print(f"  {phases[i]:>8}: {mean[i]:8.2f} us  ({100*mean[i]/total:4.1f}%)")
 c1 = clk[:, 10].mean() / FREQ * 1e6
 print(f"C1 (diag factor+inv) within C: {c1:.2f} us")
 # wall span of the whole launch from clocks
@@ -61,4 +62,4 @@ for Ee in (256, 2048):
     dd = (clk2[:, 1:10] - clk2[:, 0:9]) / 1e8 * 1e6
     m = dd.mean(0)
     print(f"E={Ee}: {dt*1e3:.3f} ms/launch; phases us: "
-          + " ".join(f"{names[i+1] if i<9 else ''}:{m[i]:.1f}" for i in range(9)))
+          + " ".join(f"{phases[i]}:{m[i]:.1f}" for i in range(9)))

@@ -37,8 +37,13 @@ def logdet_and_inv(K: torch.Tensor, force_lu: bool = False
     ``force_lu`` skips the Cholesky attempt (used when the caller already
     knows it breaks down, e.g. the HIP kernel's bad-expert fallback)."""
     if force_lu:
-        _, logabsdet = torch.linalg.slogdet(K)
-        return logabsdet, torch.linalg.inv(K)
+        # ONE LU factorization for both logdet and inverse, like the
+        # reference's logDetAndInv (dgetrf once, then dgetri)
+        LU, piv = torch.linalg.lu_factor(K)
+        logdet = torch.log(LU.diagonal(dim1=-2, dim2=-1).abs()).sum(-1)
+        eye = torch.eye(K.shape[-1], dtype=K.dtype,
+                        device=K.device).expand_as(K)
+        return logdet, torch.linalg.lu_solve(LU, piv, eye)
     L, info = torch.linalg.cholesky_ex(K)
     bad = info > 0
     if bad.any():

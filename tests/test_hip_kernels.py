@@ -240,3 +240,39 @@ def test_gpu_classifier_end_to_end(dev, ext):
              .setDevice("cuda:0")).fit(X, y)
     acc = accuracy(y, model.predict(X))
     assert acc > 0.97
+
+
+def test_fused_expert_nll_d128_vs_oracle(dev, ext):
+    """High-dimension path (config-5 shape: d=128 > old d<=64 limit)."""
+    from spark_gp_amd.kernels import (ARDRBFKernel, EyeKernel, Scalar,
+                                      compile_kernel)
+    from spark_gp_amd.ops import hip_backend, torch_backend
+    E, k, d = 8, 100, 128
+    g = torch.Generator().manual_seed(21)
+    X = torch.rand(E, k, d, generator=g).to(dev)
+    y = torch.sin(X.sum(-1)).to(dev)
+    cs = compile_kernel(1 * ARDRBFKernel(d) + Scalar(1e-3).const * EyeKernel())
+    rng = np.random.default_rng(2)
+    theta = np.concatenate([[1.1], rng.uniform(0.05, 0.3, d)])
+    nll_h, grad_h = hip_backend.nll_grad_compiled(cs, theta, X, y)
+    nll_o, grad_o = torch_backend.nll_grad_compiled(
+        cs, theta, X.double().cpu(), y.double().cpu())
+    assert nll_h == pytest.approx(nll_o, rel=3e-4)
+    np.testing.assert_allclose(grad_h, grad_o, rtol=5e-3,
+                               atol=3e-3 * np.abs(grad_o).max())
+
+
+def test_force_lu_fallback_matches(dev, ext):
+    from spark_gp_amd.kernels import (ARDRBFKernel, EyeKernel, Scalar,
+                                      compile_kernel)
+    from spark_gp_amd.ops import torch_backend
+    E, k, d = 8, 40, 4
+    g = torch.Generator().manual_seed(3)
+    X = torch.rand(E, k, d, generator=g).to(dev)
+    y = torch.sin(X.sum(-1)).to(dev)
+    cs = compile_kernel(1 * ARDRBFKernel(d) + Scalar(1e-3).const * EyeKernel())
+    theta = np.concatenate([[1.0], np.ones(d)])
+    a = torch_backend.nll_grad_compiled(cs, theta, X, y)
+    b = torch_backend.nll_grad_compiled(cs, theta, X, y, force_lu=True)
+    assert a[0] == pytest.approx(b[0], rel=1e-4)
+    np.testing.assert_allclose(a[1], b[1], rtol=1e-3, atol=1e-5)
