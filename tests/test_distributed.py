@@ -155,3 +155,44 @@ def test_distributed_fit_agrees_across_ranks_and_predicts():
     yq = np.sin(Xq.sum(-1) * 3.0)
     rmse = float(np.sqrt(np.mean((results[0] - yq) ** 2)))
     assert rmse < 0.2, rmse
+
+
+def _w_kmeans(rank, world):
+    from spark_gp_amd import KMeansActiveSetProvider
+    from spark_gp_amd.parallel.dist import get_comm
+    X, y = performance_benchmark_data(400, 3, seed=7, dtype=np.float64)
+    Xl, yl = _shard(X, y, rank, world)
+    prov = KMeansActiveSetProvider(max_iter=5)
+    centers = prov(10, torch.tensor(Xl), torch.tensor(yl), None, None, 3,
+                   get_comm())
+    return centers.numpy()
+
+
+def test_kmeans_provider_distributed_matches_single():
+    results = _spawn("_w_kmeans")
+    from spark_gp_amd import KMeansActiveSetProvider
+    from spark_gp_amd.parallel.dist import Comm
+    X, y = performance_benchmark_data(400, 3, seed=7, dtype=np.float64)
+    ref = KMeansActiveSetProvider(max_iter=5)(
+        10, torch.tensor(X), torch.tensor(y), None, None, 3, Comm()).numpy()
+    np.testing.assert_allclose(results[0], ref, atol=1e-12)
+    np.testing.assert_allclose(results[0], results[1], atol=0)
+
+
+def _w_scaling(rank, world):
+    from spark_gp_amd import StandardScaler
+    from spark_gp_amd.parallel.dist import get_comm
+    X, y = performance_benchmark_data(400, 3, seed=7, dtype=np.float64)
+    Xl, _ = _shard(X, y, rank, world)
+    s = StandardScaler().fit(Xl, get_comm())
+    return np.concatenate([s.mean, s.scale])
+
+
+def test_scaling_distributed_matches_single():
+    results = _spawn("_w_scaling")
+    from spark_gp_amd import StandardScaler
+    X, y = performance_benchmark_data(400, 3, seed=7, dtype=np.float64)
+    s = StandardScaler().fit(X)
+    ref = np.concatenate([s.mean, s.scale])
+    np.testing.assert_allclose(results[0], ref, rtol=1e-12)
+    np.testing.assert_allclose(results[1], ref, rtol=1e-12)

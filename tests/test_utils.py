@@ -139,3 +139,38 @@ def test_memoized_objective_caches():
     assert len(calls) == 1 and a[0] == b[0]
     memo(np.array([3.0, 4.0]))
     assert len(calls) == 2
+
+
+def test_classifier_model_save_load_roundtrip(tmp_path):
+    from spark_gp_amd import GaussianProcessClassifier
+    rng = np.random.default_rng(0)
+    n = 120
+    X = np.concatenate([rng.normal(-2, 0.7, (n // 2, 2)),
+                        rng.normal(2, 0.7, (n // 2, 2))])
+    y = np.concatenate([np.zeros(n // 2), np.ones(n // 2)])
+    model = (GaussianProcessClassifier()
+             .setKernel(lambda: 1 * RBFKernel(1.0, 1e-3, 10))
+             .setDatasetSizeForExpert(40)
+             .setActiveSetSize(30)
+             .setSigma2(1e-3)
+             .setMaxIter(15)
+             .setSeed(7)
+             .setDevice("cpu")).fit(X, y)
+    p1 = model.predict_proba(X[:20])
+    save_model(model, str(tmp_path / "clf"))
+    loaded = load_model(str(tmp_path / "clf"))
+    assert type(loaded).__name__ == "GaussianProcessClassificationModel"
+    p2 = loaded.predict_proba(X[:20])
+    np.testing.assert_allclose(p1, p2, rtol=1e-10)
+
+
+def test_integrator_averaged_proba_single_matches_batch():
+    integ = Integrator(32)
+    f = np.array([0.5, -1.0])
+    v = np.array([0.3, 1.5])
+    batch = integ.expected_of_function_of_normal_batch(
+        f, v, lambda z: 1.0 / (1.0 + np.exp(-z)))
+    for i in range(2):
+        single = integ.expected_of_function_of_normal(
+            f[i], v[i], lambda z: 1.0 / (1.0 + np.exp(-z)))
+        assert single == pytest.approx(batch[i], rel=1e-12)
