@@ -386,7 +386,7 @@ PH(2);
         __syncwarp();
       }
       __syncwarp();
-    } else if (J > 0) {    } else if (J > 0) {
+    } else if (J > 0) {
       // waves 1-7: previous panel's update to (a) the panel rows of
       // column-block J and (b) the remaining trailing triangle
       const int t0r = jb + bs;

@@ -40,6 +40,12 @@ def test_fused_expert_nll_ard_vs_oracle(dev, ext):
     rng = np.random.default_rng(1)
     theta = np.concatenate([[1.3], rng.uniform(0.5, 2.0, d)])
 
+    # the HIP kernel itself must handle every expert (a silent full
+    # fallback to the torch path would make this test vacuous)
+    scale = torch.as_tensor(theta[1:], dtype=torch.float32, device=dev)
+    *_, bad = ext.fused_expert_nll(X, y, scale, float(theta[0]), 1e-3)
+    assert int(bad.sum()) == 0, "experts fell back: kernel not exercised"
+
     nll_h, grad_h = hip_backend.nll_grad_compiled(cs, theta, X, y)
     # fp64 oracle on the same data
     nll_o, grad_o = torch_backend.nll_grad_compiled(
