@@ -84,6 +84,16 @@ def nll_grad_generic(kernel: Kernel, theta: np.ndarray,
 def laplace_nll_grad(kernel: Kernel, theta: np.ndarray, X: torch.Tensor,
                      y: torch.Tensor, f: torch.Tensor, tol: float,
                      max_newton_iter: int = 200):
+    if X.is_cuda and not _force_torch():
+        from ..kernels.compiled import compile_kernel
+        cs = compile_kernel(kernel)
+        hip = _load_hip()
+        if cs is not None and                 _require_hip_or_fallback("laplace_nll_grad") and                 hip.supports_laplace(cs, X):
+            # fused Newton loop runs each expert to convergence on the GPU
+            # (updates f in place); the torch pass below then converges in
+            # 2-3 cheap iterations and computes the Algorithm 5.1 evidence
+            # with the reference's exact semantics.
+            hip.laplace_newton(cs, theta, X, y, f, tol, max_newton_iter)
     return torch_backend.laplace_nll_grad(kernel, theta, X, y, f, tol,
                                           max_newton_iter)
 

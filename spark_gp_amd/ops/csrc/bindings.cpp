@@ -25,6 +25,12 @@ extern "C" hipError_t launch_colsum_gemv(const void* Kc, const float* y,
                                          int c, int m, double* Ky,
                                          hipStream_t stream);
 
+extern "C" hipError_t launch_fused_laplace_newton(
+    const float* X, const float* y, float* f, const float* scale, float amp,
+    float noise, int E, int k, int d, double tol, int max_newton,
+    double* out_psi, double* out_sumlogl, int* out_iters, int* out_bad,
+    hipStream_t stream, size_t* lds_used);
+
 namespace {
 
 void check_hip(hipError_t err, const char* what) {
@@ -152,7 +158,55 @@ void colsum_gemv_acc(torch::Tensor Kc, torch::Tensor y, torch::Tensor Ky) {
             "colsum_gemv");
 }
 
+// Runs the per-expert Laplace Newton loop in place on f.
+// Returns (psi[E] f64, sumlogl[E] f64, iters[E] i32, bad[E] i32).
+std::vector<torch::Tensor> fused_laplace_newton(torch::Tensor X,
+                                                torch::Tensor y,
+                                                torch::Tensor f,
+                                                torch::Tensor scale,
+                                                double amp, double noise,
+                                                double tol,
+                                                int64_t max_newton) {
+  TORCH_CHECK(X.is_cuda() && X.dtype() == torch::kFloat32 && X.dim() == 3);
+  TORCH_CHECK(y.is_cuda() && y.dtype() == torch::kFloat32 && y.dim() == 2);
+  TORCH_CHECK(f.is_cuda() && f.dtype() == torch::kFloat32 && f.dim() == 2
+              && f.is_contiguous(), "f must be contiguous fp32 (updated "
+              "in place)");
+  auto Xc = X.contiguous();
+  auto yc = y.contiguous();
+  auto sc = scale.contiguous();
+  const int E = X.size(0), k = X.size(1), d = X.size(2);
+  TORCH_CHECK(k <= 128 && d <= k,
+              "fused_laplace_newton requires k<=128, d<=k");
+  auto opts64 = torch::TensorOptions().dtype(torch::kFloat64).device(X.device());
+  auto opts32i = torch::TensorOptions().dtype(torch::kInt32).device(X.device());
+  auto psi = torch::empty({E}, opts64);
+  auto sll = torch::empty({E}, opts64);
+  auto iters = torch::empty({E}, opts32i);
+  auto bad = torch::empty({E}, opts32i);
+  size_t lds = 0;
+  check_hip(launch_fused_laplace_newton(
+                Xc.data_ptr<float>(), yc.data_ptr<float>(),
+                f.data_ptr<float>(), sc.data_ptr<float>(), (float)amp,
+                (float)noise, E, k, d, tol, (int)max_newton,
+                psi.data_ptr<double>(), sll.data_ptr<double>(),
+                iters.data_ptr<int>(), bad.data_ptr<int>(), current_stream(),
+                &lds),
+            "fused_laplace_newton");
+  return {psi, sll, iters, bad};
+}
+
+bool fused_laplace_newton_supported(int64_t k, int64_t d) {
+  if (k > 128 || d > k || k < 1) return false;
+  const int64_t tsz = std::max<int64_t>(k * 33, 448);
+  int64_t bytes = 80 + 4 * (2 * k * (k + 1) + tsz + 8 * k + d) + 16;
+  return bytes <= 160 * 1024;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("fused_laplace_newton", &fused_laplace_newton,
+          "per-expert Laplace Newton loop to convergence (CDNA4)");
+  mod.def("fused_laplace_newton_supported", &fused_laplace_newton_supported);
   mod.def("fused_expert_nll", &fused_expert_nll,
           "fused per-expert BCM nll+gradient primitives (CDNA4)");
   mod.def("fused_expert_nll_profile", &fused_expert_nll_impl,

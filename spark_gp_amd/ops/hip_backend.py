@@ -88,6 +88,37 @@ def nll_grad_compiled(cs: CompiledKernel, theta: np.ndarray,
 
 
 # ---------------------------------------------------------------------------
+# Laplace Newton pre-pass (GPC)
+# ---------------------------------------------------------------------------
+
+def supports_laplace(cs: CompiledKernel, X: torch.Tensor) -> bool:
+    if cs is None or cs.base not in ("ard", "rbf"):
+        return False
+    if X.dtype != torch.float32 or X.dim() != 3:
+        return False
+    E, k, d = X.shape
+    return bool(ext.fused_laplace_newton_supported(k, d))
+
+
+def laplace_newton(cs: CompiledKernel, theta: np.ndarray, X: torch.Tensor,
+                   y: torch.Tensor, f: torch.Tensor, tol: float,
+                   max_newton: int) -> int:
+    """Run the fused per-expert Newton loop to convergence, updating the
+    latent ``f`` IN PLACE.  Returns the number of experts the fp32 kernel
+    could not handle (their f is left unchanged; the torch evidence pass
+    converges them from their warm state)."""
+    C = cs.amp(theta)
+    nu = cs.noise(theta)
+    scale = _scale_vector(cs, theta, X.shape[-1], X.device)
+    if not f.is_contiguous():
+        raise ValueError("latent f must be contiguous")
+    psi, sll, iters, bad = ext.fused_laplace_newton(
+        X, y.to(torch.float32), f, scale, float(C), float(nu), float(tol),
+        int(max_newton))
+    return int((bad != 0).sum())
+
+
+# ---------------------------------------------------------------------------
 # PPA path
 # ---------------------------------------------------------------------------
 
