@@ -125,7 +125,7 @@ __device__ inline void chol_invert_lower(float* Abuf, float* Tbuf,
 #pragma unroll
           for (int ss = 0; ss < 8; ++ss) {
             const float piv = m[ss][ss];
-            if (ss < sbs) {
+            if (ss < sbs && ok) {     // classify only the FIRST failure
               if (!isfinite(piv)) { ok = false; nonfin = true; }
               else if (!(piv > 0.f)) ok = false;
               else ldet += (double)__logf(piv);
@@ -166,10 +166,14 @@ __device__ inline void chol_invert_lower(float* Abuf, float* Tbuf,
 #pragma unroll
             for (int j = 0; j < 8; ++j)
               Vq[q * 64 + i * 8 + j] = v8[i][j];
-          if (!ok) *bad = nonfin ? 2 : 1;   // 2: non-finite iterate
-          else misc[0] += ldet;
+          if (!ok) {
+            if (*bad == 0) *bad = nonfin ? 2 : 1;  // sticky: first cause
+          } else {
+            misc[0] += ldet;
+          }
         }
         __syncwarp();
+        if (*bad) break;               // wave-uniform: skip garbage blocks
         // panel rows within the 32-block: P = A * Vq^T; then trailing
         const int p0 = qb + sbs;       // first panel row (local)
         const int pr = bs - p0;        // panel rows
