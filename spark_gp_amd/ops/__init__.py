@@ -93,7 +93,13 @@ def laplace_nll_grad(kernel: Kernel, theta: np.ndarray, X: torch.Tensor,
             # (updates f in place); the torch pass below then converges in
             # 2-3 cheap iterations and computes the Algorithm 5.1 evidence
             # with the reference's exact semantics.
-            hip.laplace_newton(cs, theta, X, y, f, tol, max_newton_iter)
+            n_bad = hip.laplace_newton(cs, theta, X, y, f, tol,
+                                       max_newton_iter)
+            # evidence at the converged f; only if some experts fell back
+            # does the torch Newton loop still run (from their warm f)
+            return torch_backend.laplace_nll_grad(
+                kernel, theta, X, y, f, tol, max_newton_iter,
+                newton=(n_bad > 0))
     return torch_backend.laplace_nll_grad(kernel, theta, X, y, f, tol,
                                           max_newton_iter)
 

@@ -158,8 +158,8 @@ def nll_grad_generic(kernel: Kernel, theta: np.ndarray,
 
 def laplace_nll_grad(kernel: Kernel, theta: np.ndarray,
                      X: torch.Tensor, y: torch.Tensor, f: torch.Tensor,
-                     tol: float, max_newton_iter: int = 200
-                     ) -> Tuple[float, np.ndarray]:
+                     tol: float, max_newton_iter: int = 200,
+                     newton: bool = True) -> Tuple[float, np.ndarray]:
     """Batched Newton iteration (R&W Algorithm 3.1 with step halving) +
     Algorithm 5.1 evidence/gradient, mirroring
     ``classification/GaussianProcessClassifier.scala:74-129``.
@@ -188,6 +188,26 @@ def laplace_nll_grad(kernel: Kernel, theta: np.ndarray,
     sqw_out = torch.empty_like(f)
 
     active = torch.ones(E, dtype=torch.bool, device=dev)
+    if not newton:
+        # f already converged (fused HIP Newton pre-pass): evaluate the
+        # exit-state quantities of Algorithm 3.1 once at the current f
+        pi = torch.sigmoid(f)
+        w = pi * (1.0 - pi)
+        sqw = torch.sqrt(w)
+        B = eyek + sqw.unsqueeze(-1) * K * sqw.unsqueeze(-2)
+        L_out = torch.linalg.cholesky(B)
+        grad_logp = y - pi
+        b = w * f + grad_logp
+        Kb = (K @ b.unsqueeze(-1)).squeeze(-1)
+        v = torch.cholesky_solve((sqw * Kb).unsqueeze(-1), L_out).squeeze(-1)
+        a_out = b - sqw * v
+        pi_out, sqw_out = pi, sqw
+        fc = (K @ a_out.unsqueeze(-1)).squeeze(-1)
+        new_obj = (-0.5 * (a_out * fc).sum(-1).double()
+                   + torch.nn.functional.logsigmoid(
+                       (2.0 * y - 1.0) * fc).double().sum(-1))
+        active = torch.zeros(E, dtype=torch.bool, device=dev)
+
     it = 0
     while bool(active.any()) and it < max_newton_iter:
         it += 1
