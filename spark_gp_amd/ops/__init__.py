@@ -95,11 +95,14 @@ def laplace_nll_grad(kernel: Kernel, theta: np.ndarray, X: torch.Tensor,
             # with the reference's exact semantics.
             n_bad = hip.laplace_newton(cs, theta, X, y, f, tol,
                                        max_newton_iter)
-            # evidence at the converged f; only if some experts fell back
-            # does the torch Newton loop still run (from their warm f)
+            if n_bad == 0:
+                # Algorithm 5.1 at the converged f via the contraction
+                # form — no [E, p, k, k] derivative tensor
+                return torch_backend.laplace_evidence_compiled(
+                    cs, theta, X, y, f)
+            # some experts fell back: torch Newton finishes them warm
             return torch_backend.laplace_nll_grad(
-                kernel, theta, X, y, f, tol, max_newton_iter,
-                newton=(n_bad > 0))
+                kernel, theta, X, y, f, tol, max_newton_iter)
     return torch_backend.laplace_nll_grad(kernel, theta, X, y, f, tol,
                                           max_newton_iter)
 

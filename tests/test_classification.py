@@ -162,3 +162,27 @@ def test_one_vs_rest_iris():
     model = ovr.fit(X, y)
     acc = accuracy(y, model.predict(X))
     assert acc > 0.9
+
+
+def test_laplace_evidence_compiled_matches_generic():
+    """K11 contraction-form evidence (no [E,p,k,k] tensor) vs the generic
+    materialized-derivative path, at a converged latent f."""
+    from spark_gp_amd.kernels import compile_kernel, WhiteNoiseKernel
+    g = torch.Generator().manual_seed(0)
+    for factory, theta in [
+        (lambda: 1 * ARDRBFKernel(3) + Scalar(1e-2).const * EyeKernel(),
+         np.array([1.1, 0.8, 1.2, 0.6])),
+        (lambda: 1 * RBFKernel(0.7) + WhiteNoiseKernel(0.1, 0, 1)
+         + Scalar(1e-3).const * EyeKernel(),
+         np.array([0.9, 0.75, 0.12])),
+    ]:
+        X = torch.randn(4, 18, 3, generator=g, dtype=TD)
+        y = (X.sum(-1) > 0).double()
+        kern = factory()
+        cs = compile_kernel(kern)
+        f = torch.zeros(4, 18, dtype=TD)
+        nll_ref, grad_ref = tb.laplace_nll_grad(kern, theta, X, y, f, 1e-10)
+        nll_c, grad_c = tb.laplace_evidence_compiled(cs, theta, X, y, f)
+        assert nll_c == pytest.approx(nll_ref, rel=1e-6)
+        np.testing.assert_allclose(grad_c, grad_ref, rtol=1e-5,
+                                   atol=1e-6 * np.abs(grad_ref).max())
