@@ -13,8 +13,6 @@ Mirrors ``commons/ActiveSetProvider.scala``:
 from __future__ import annotations
 
 import math
-from typing import Optional
-
 import numpy as np
 import torch
 

@@ -15,7 +15,6 @@ MI355X-first deviations (recorded in SURVEY.md §7):
 
 from __future__ import annotations
 
-import math
 import time
 from typing import Callable, List, Optional, Tuple
 
@@ -24,10 +23,9 @@ import torch
 
 from ..active_set import ActiveSetProvider, RandomActiveSetProvider
 from ..kernels.base import EyeKernel, Kernel, Scalar
-from ..kernels.compiled import compile_kernel
 from ..kernels.rbf import RBFKernel
-from ..optimize import MemoizedObjective, lbfgsb
-from ..parallel.dist import Comm, get_comm
+from ..optimize import lbfgsb
+from ..parallel.dist import Comm
 from ..ppa import accumulate_ppa_stats, magic_vector_matrix
 from ..utils.instrumentation import Instrumentation
 from .predictor import GaussianProjectedProcessRawPredictor

@@ -10,7 +10,7 @@ L-BFGS evaluations, and the PPA fit on the latent f as regression target
 
 from __future__ import annotations
 
-from typing import List, Optional, Tuple
+from typing import Optional, Tuple
 
 import numpy as np
 import torch

@@ -15,7 +15,6 @@ import os
 from typing import Dict
 
 import numpy as np
-import torch
 
 from ..kernels import (ARDRBFKernel, ConstantTimesKernel, EyeKernel, Kernel,
                        RBFKernel, SumOfKernels, TrainableScalarTimesKernel)

@@ -14,7 +14,7 @@ generic materialized-derivative path is used.
 
 from __future__ import annotations
 
-from typing import List, Optional, Tuple
+from typing import Optional, Tuple
 
 import numpy as np
 import torch

@@ -9,7 +9,7 @@ interchangeable and unit-diffable.
 from __future__ import annotations
 
 import math
-from typing import Optional, Tuple
+from typing import Tuple
 
 import numpy as np
 import torch

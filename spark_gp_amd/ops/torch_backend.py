@@ -17,7 +17,7 @@ done in float64 regardless of the compute dtype.
 from __future__ import annotations
 
 import math
-from typing import Optional, Tuple
+from typing import Tuple
 
 import numpy as np
 import torch

@@ -5,7 +5,6 @@ split, RMSE / accuracy evaluators and a one-vs-rest multiclass wrapper.
 
 from __future__ import annotations
 
-import copy
 from typing import Callable, List
 
 import numpy as np
