@@ -88,7 +88,9 @@ def laplace_nll_grad(kernel: Kernel, theta: np.ndarray, X: torch.Tensor,
         from ..kernels.compiled import compile_kernel
         cs = compile_kernel(kernel)
         hip = _load_hip()
-        if cs is not None and                 _require_hip_or_fallback("laplace_nll_grad") and                 hip.supports_laplace(cs, X):
+        if (cs is not None
+                and _require_hip_or_fallback("laplace_nll_grad")
+                and hip.supports_laplace(cs, X)):
             # fused Newton loop runs each expert to convergence on the GPU
             # (updates f in place); the torch pass below then converges in
             # 2-3 cheap iterations and computes the Algorithm 5.1 evidence

@@ -179,8 +179,6 @@ __device__ inline void chol_invert_lower(float* Abuf, float* Tbuf,
         const int pr = bs - p0;        // panel rows
         if (pr > 0) {
           // panel: (pr x sbs) elements, lanes parallel
-          float pv[1];
-          (void)pv;
           for (int f = lane; f < pr * 8; f += 64) {
             const int r = f >> 3, c = f & 7;
             if (c >= sbs) continue;
