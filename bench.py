@@ -110,9 +110,15 @@ def main():
     # the first fits on a cold GPU under-reports steady throughput by 3-4x).
     t_w = time.perf_counter()
     w = 0
-    while w < args.warmup or (device.type == "cuda"
-                              and time.perf_counter() - t_w
-                              < args.min_warmup_seconds):
+    while True:
+        cont = w < args.warmup or (device.type == "cuda"
+                                   and time.perf_counter() - t_w
+                                   < args.min_warmup_seconds)
+        # the continue-decision must be collective: a fit contains
+        # allreduces, so every rank must run the same number of warmup fits
+        cont = comm.allreduce_scalar(1.0 if cont else 0.0, op="max") > 0.5
+        if not cont:
+            break
         model = make_gp().fit(Xt, yt)
         w += 1
 
