@@ -17,7 +17,8 @@ from typing import Dict
 import numpy as np
 
 from ..kernels import (ARDRBFKernel, ConstantTimesKernel, EyeKernel, Kernel,
-                       RBFKernel, SumOfKernels, TrainableScalarTimesKernel)
+                       Matern32Kernel, Matern52Kernel, RBFKernel, SumOfKernels,
+                       TrainableScalarTimesKernel)
 from .classification import GaussianProcessClassificationModel
 from .predictor import GaussianProjectedProcessRawPredictor
 from .regression import GaussianProcessRegressionModel
@@ -51,6 +52,9 @@ def kernel_to_spec(k: Kernel) -> Dict:
     if isinstance(k, RBFKernel):
         return {"op": "rbf", "sigma": k.sigma, "lower": _enc(k.lower),
                 "upper": _enc(k.upper)}
+    if isinstance(k, (Matern32Kernel, Matern52Kernel)):
+        return {"op": "m32" if isinstance(k, Matern32Kernel) else "m52",
+                "l": k.l, "lower": _enc(k.lower), "upper": _enc(k.upper)}
     if isinstance(k, ARDRBFKernel):
         return {"op": "ard", "beta": k.beta.tolist(),
                 "lower": [_enc(float(v)) for v in k.lower],
@@ -73,6 +77,9 @@ def kernel_from_spec(spec: Dict) -> Kernel:
         return EyeKernel()
     if op == "rbf":
         return RBFKernel(spec["sigma"], _dec(spec["lower"]), _dec(spec["upper"]))
+    if op in ("m32", "m52"):
+        cls = Matern32Kernel if op == "m32" else Matern52Kernel
+        return cls(spec["l"], _dec(spec["lower"]), _dec(spec["upper"]))
     if op == "ard":
         return ARDRBFKernel(np.array(spec["beta"]),
                             beta=1.0,
