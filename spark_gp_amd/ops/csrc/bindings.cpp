@@ -14,12 +14,12 @@ extern "C" hipError_t launch_fused_expert_nll(
 
 extern "C" hipError_t launch_cross_kernel_tile(
     const float* X, const float* A, const float* s2v, float amp, int c,
-    int m, int d, void* out, void* out_lo, int out_is_bf16,
-    hipStream_t stream);
+    int m, int d, void* out, void* out_lo, void* out_t, void* out_lo_t,
+    int out_is_bf16, hipStream_t stream);
 
-extern "C" hipError_t launch_syrk_bf16(const void* Kc, const void* Kl,
-                                       int c, int m, int split_k, float* KK,
-                                       hipStream_t stream);
+extern "C" hipError_t launch_syrk_bf16(const void* KcT, const void* KlT,
+                                       int c, int m, int cpitch, int split_k,
+                                       float* KK, hipStream_t stream);
 
 extern "C" hipError_t launch_colsum_gemv(const void* Kc, const float* y,
                                          int c, int m, double* Ky,
@@ -103,11 +103,13 @@ bool fused_expert_nll_supported(int64_t k, int64_t d) {
 
 std::vector<torch::Tensor> cross_kernel_tile(torch::Tensor X, torch::Tensor A,
                                              torch::Tensor s2v, double amp,
-                                             bool bf16_out, bool hilo) {
+                                             bool bf16_out, bool hilo,
+                                             bool want_t) {
   TORCH_CHECK(X.is_cuda() && X.dtype() == torch::kFloat32 && X.dim() == 2);
   TORCH_CHECK(A.is_cuda() && A.dtype() == torch::kFloat32 && A.dim() == 2);
   TORCH_CHECK(X.size(1) == A.size(1), "feature dims differ");
   TORCH_CHECK(!hilo || bf16_out, "hilo requires bf16 output");
+  TORCH_CHECK(!want_t || hilo, "transposed outputs require hilo bf16 mode");
   auto Xc = X.contiguous();
   auto Ac = A.contiguous();
   auto sc = s2v.contiguous();
@@ -116,33 +118,45 @@ std::vector<torch::Tensor> cross_kernel_tile(torch::Tensor X, torch::Tensor A,
                   .dtype(bf16_out ? torch::kBFloat16 : torch::kFloat32)
                   .device(X.device());
   auto out = torch::empty({c, m}, opts);
-  torch::Tensor lo;
+  torch::Tensor lo, outT, loT;
   if (hilo) lo = torch::empty({c, m}, opts);
+  if (want_t) {
+    outT = torch::empty({m, c}, opts);
+    loT = torch::empty({m, c}, opts);
+  }
   check_hip(launch_cross_kernel_tile(Xc.data_ptr<float>(), Ac.data_ptr<float>(),
                                      sc.data_ptr<float>(), (float)amp, c, m, d,
                                      out.data_ptr(),
                                      hilo ? lo.data_ptr() : nullptr,
+                                     want_t ? outT.data_ptr() : nullptr,
+                                     want_t ? loT.data_ptr() : nullptr,
                                      bf16_out ? 1 : 0, current_stream()),
             "cross_kernel_tile");
+  if (want_t) return {out, lo, outT, loT};
   if (hilo) return {out, lo};
   return {out};
 }
 
-void syrk_bf16_acc(torch::Tensor Kc, c10::optional<torch::Tensor> Kl,
+// KcT/KlT are the TRANSPOSED kernel blocks [m, c] (k along the contiguous
+// axis; see syrk_bf16_kernel).  KK [m, m] is accumulated in place.
+void syrk_bf16_acc(torch::Tensor KcT, c10::optional<torch::Tensor> KlT,
                    torch::Tensor KK, int64_t split_k) {
-  TORCH_CHECK(Kc.is_cuda() && Kc.dtype() == torch::kBFloat16 && Kc.dim() == 2);
+  TORCH_CHECK(KcT.is_cuda() && KcT.dtype() == torch::kBFloat16 &&
+              KcT.dim() == 2);
   TORCH_CHECK(KK.is_cuda() && KK.dtype() == torch::kFloat32 && KK.dim() == 2);
-  auto Kcc = Kc.contiguous();
-  const int c = Kc.size(0), m = Kc.size(1);
-  TORCH_CHECK(KK.size(0) == m && KK.size(1) == m);
+  auto Kcc = KcT.contiguous();
+  const int m = KcT.size(0), c = KcT.size(1);
+  TORCH_CHECK(KK.size(0) == m && KK.size(1) == m,
+              "KK must be [m, m] with m = KcT.size(0) (transposed operand)");
   torch::Tensor Klc;
-  if (Kl.has_value()) {
-    TORCH_CHECK(Kl->sizes() == Kc.sizes() && Kl->dtype() == torch::kBFloat16);
-    Klc = Kl->contiguous();
+  if (KlT.has_value()) {
+    TORCH_CHECK(KlT->sizes() == KcT.sizes() &&
+                KlT->dtype() == torch::kBFloat16);
+    Klc = KlT->contiguous();
   }
   check_hip(launch_syrk_bf16(Kcc.data_ptr(),
-                             Kl.has_value() ? Klc.data_ptr() : nullptr, c, m,
-                             (int)split_k, KK.data_ptr<float>(),
+                             KlT.has_value() ? Klc.data_ptr() : nullptr, c, m,
+                             c, (int)split_k, KK.data_ptr<float>(),
                              current_stream()),
             "syrk_bf16");
 }
@@ -213,9 +227,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           "same, with per-phase wall_clock64 boundaries appended");
   mod.def("fused_expert_nll_supported", &fused_expert_nll_supported);
   mod.def("cross_kernel_tile", &cross_kernel_tile,
-          "rectangular RBF/ARD kernel block (CDNA4)");
+          "rectangular RBF/ARD kernel block (CDNA4)",
+          pybind11::arg("X"), pybind11::arg("A"), pybind11::arg("s2v"),
+          pybind11::arg("amp"), pybind11::arg("bf16_out"),
+          pybind11::arg("hilo"), pybind11::arg("want_t") = false);
   mod.def("syrk_bf16_acc", &syrk_bf16_acc,
-          "KK += Kc^T Kc, bf16 MFMA, fp32 accumulate (CDNA4)");
+          "KK += K K^T of the transposed block KcT [m, c]; bf16 MFMA, "
+          "fp32 accumulate (CDNA4)");
   mod.def("colsum_gemv_acc", &colsum_gemv_acc,
           "Ky += Kc^T y, fp64 accumulate (CDNA4)");
 }

@@ -37,6 +37,8 @@ cross_kernel_tile_kernel(const float* __restrict__ X,    // [c, d]
                          bf16* __restrict__ out_bf,      // [c, m] or null
                          bf16* __restrict__ out_lo,      // [c, m] or null:
                                                          // residual v - hi
+                         bf16* __restrict__ out_bfT,     // [m, c] or null
+                         bf16* __restrict__ out_loT,     // [m, c] or null
                          float* __restrict__ out_f32) {  // [c, m] or null
   __shared__ float xs[CK_TILE][CK_DBLK + 1];
   __shared__ float as[CK_TILE][CK_DBLK + 1];
@@ -86,6 +88,19 @@ cross_kernel_tile_kernel(const float* __restrict__ X,    // [c, d]
     __syncthreads();
   }
 
+  // finalize the 8x8 micro-tile once; hi/lo split: hi + lo carries ~16
+  // mantissa bits, so the SYRK's input-quantization error drops to fp32 class
+  bf16 hv[8][8], lv[8][8];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float v = amp * __expf(-acc[i][j]);
+      acc[i][j] = v;
+      hv[i][j] = (bf16)v;
+      lv[i][j] = (bf16)(v - (float)hv[i][j]);
+    }
+
 #pragma unroll
   for (int i = 0; i < 8; ++i) {
     const int gr = row0 + tr + i;
@@ -94,15 +109,37 @@ cross_kernel_tile_kernel(const float* __restrict__ X,    // [c, d]
     for (int j = 0; j < 8; ++j) {
       const int gc = col0 + tc + j;
       if (gc >= m) continue;
-      const float v = amp * __expf(-acc[i][j]);
       if (out_bf) {
-        const bf16 hi = (bf16)v;
-        out_bf[(size_t)gr * m + gc] = hi;
-        // two-term bf16 split: hi + lo carries ~16 mantissa bits, so the
-        // SYRK's input-quantization error drops to fp32 class
-        if (out_lo) out_lo[(size_t)gr * m + gc] = (bf16)(v - (float)hi);
+        out_bf[(size_t)gr * m + gc] = hv[i][j];
+        if (out_lo) out_lo[(size_t)gr * m + gc] = lv[i][j];
       } else {
-        out_f32[(size_t)gr * m + gc] = v;
+        out_f32[(size_t)gr * m + gc] = acc[i][j];
+      }
+    }
+  }
+
+  if (out_bfT) {
+    // transposed copies [m, c] so the SYRK can stage k-contiguous: per
+    // output column j, the 8 row values are contiguous -> one b128 store
+    const int gr0 = row0 + tr;
+    const bool vec = (c % 8 == 0) && (gr0 + 8 <= c);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int gc = col0 + tc + j;
+      if (gc >= m) continue;
+      bf16 th[8], tl[8];
+#pragma unroll
+      for (int i = 0; i < 8; ++i) { th[i] = hv[i][j]; tl[i] = lv[i][j]; }
+      if (vec) {
+        *(uint4*)&out_bfT[(size_t)gc * c + gr0] = *(uint4*)th;
+        if (out_loT) *(uint4*)&out_loT[(size_t)gc * c + gr0] = *(uint4*)tl;
+      } else {
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          if (gr0 + i >= c) break;
+          out_bfT[(size_t)gc * c + gr0 + i] = th[i];
+          if (out_loT) out_loT[(size_t)gc * c + gr0 + i] = tl[i];
+        }
       }
     }
   }
@@ -111,37 +148,46 @@ cross_kernel_tile_kernel(const float* __restrict__ X,    // [c, d]
 // ---------------------------------------------------------------------------
 // syrk_bf16: KK[m, m] += Kc^T Kc, Kc [c, m] bf16 row-major
 // ---------------------------------------------------------------------------
-// 256-thread block = 4 waves in a 2x2 grid; 128x128 output tile per block;
-// each wave computes a 64x64 sub-tile as 4x4 mfma_f32_16x16x32_bf16 tiles.
-// K-loop: BK=32 rows of Kc staged k-major in LDS: lt[i][kk] / rt[j][kk]
-// with row stride 40 bf16 (80 B, 16-B aligned, conflict-free for the
-// 16-lane groups of ds_read_b128 — banks hit 20*i mod 64, a full
-// permutation).  Fragment loads are single ds_read_b128 each.
+// 512-thread block = 8 waves in a 4x2 grid; 256x256 output tile per block
+// (the largest tile the LDS+register budget allows: per-CU VMEM return
+// throughput (~10 B/cyc/CU HBM-bound) is the wall, and bytes/flop scales
+// as (M+N)/(M*N), so 256^2 halves traffic vs 128^2); each wave computes a
+// 64x128 sub-tile as 4x8 mfma_f32_16x16x32_bf16 tiles.  K-loop: BK=32
+// rows staged k-major in LDS; split-K over the chunk rows for occupancy.
 
 #define SY_BK 32
-#define SY_STR 136    // row-major [SY_BK][SY_STR] bf16; 272-B rows (16-B
-                      // aligned so staging is plain b128 writes)
+#define SY_CT 256     // output tile edge (per block)
+#define SY_WG 512
+#define SY_STR 40     // k-major [SY_CT][SY_STR] bf16: 32 k + 8 pad = 80-B
+                      // rows.  80 B x 16 consecutive lanes covers all 64
+                      // LDS banks exactly (20c mod 64 is a permutation of
+                      // 4-bank blocks), so staging b128 writes AND fragment
+                      // b128 reads are both conflict-free.  4 operands x
+                      // 20 KB = 80 KB -> one 8-wave block/CU (2 waves/SIMD).
 
 template <bool HILO>
-__global__ void __launch_bounds__(256)
-syrk_bf16_kernel(const bf16* __restrict__ Kc,   // [c, m] hi part
-                 const bf16* __restrict__ Kl,   // [c, m] lo part (HILO only)
-                 const int c, const int m, const int ntile,
+__global__ void __launch_bounds__(SY_WG, 2)  // force 2 waves/SIMD occupancy
+syrk_bf16_kernel(const bf16* __restrict__ KcT,  // [m, cpitch] hi^T
+                 const bf16* __restrict__ KlT,  // [m, cpitch] lo^T (HILO)
+                 const int c, const int m, const int cpitch, const int ntile,
                  const int split_k,
                  float* __restrict__ KK) {      // [m, m] accumulated
   // with HILO: KK += hi^T hi + hi^T lo + lo^T hi  (lo^T lo ~ 2^-32, dropped)
-  // Tiles stage ROW-major ([k][col]): global loads are coalesced 16-B
-  // vectors and LDS writes are conflict-free b128; the k-strided MFMA
-  // fragment gathers are scalar b16 reads (2-way worst case).
-  __shared__ __align__(16) bf16 lt[SY_BK * SY_STR];
-  __shared__ __align__(16) bf16 rt[SY_BK * SY_STR];
-  __shared__ __align__(16) bf16 ltl[HILO ? SY_BK * SY_STR : 1];
-  __shared__ __align__(16) bf16 rtl[HILO ? SY_BK * SY_STR : 1];
+  // Operands arrive TRANSPOSED ([m, c], written for free by the cross
+  // kernel's register tile), so k runs along the contiguous axis: staging
+  // is one uint4 global load (8 lanes = one full 128-B line) + one
+  // conflict-free b128 LDS write per thread, and every MFMA fragment is
+  // one b128 LDS read.  The old [c, m]-operand versions were LDS- or
+  // VMEM-instruction bound (8 scalar ops per fragment either way).
+  __shared__ __align__(16) bf16 lt[SY_CT * SY_STR];
+  __shared__ __align__(16) bf16 rt[SY_CT * SY_STR];
+  __shared__ __align__(16) bf16 ltl[HILO ? SY_CT * SY_STR : 1];
+  __shared__ __align__(16) bf16 rtl[HILO ? SY_CT * SY_STR : 1];
 
   const int tile = blockIdx.x;
   const int ti = tile / ntile, tj = tile % ntile;
   if (tj < ti) return;                     // upper-triangle tiles only
-  const int i0 = ti * 128, j0 = tj * 128;
+  const int i0 = ti * SY_CT, j0 = tj * SY_CT;
   const int slice = blockIdx.y;
   const int kblocks = (c + SY_BK - 1) / SY_BK;
   const int per = (kblocks + split_k - 1) / split_k;
@@ -150,92 +196,109 @@ syrk_bf16_kernel(const bf16* __restrict__ Kc,   // [c, m] hi part
   if (kb0 >= kb1) return;
 
   const int tid = threadIdx.x;
-  const int wave = tid >> 6;              // 0..3
+  const int wave = tid >> 6;              // 0..7
   const int lane = tid & 63;
   const int wr = (wave >> 1) * 64;        // wave row offset in tile
-  const int wc = (wave & 1) * 64;         // wave col offset in tile
+  const int wc = (wave & 1) * 128;        // wave col offset in tile
 
-  f32x4 acc[4][4];
+  f32x4 acc[4][8];
 #pragma unroll
   for (int a = 0; a < 4; ++a)
 #pragma unroll
-    for (int b = 0; b < 4; ++b) acc[a][b] = {0.f, 0.f, 0.f, 0.f};
+    for (int b = 0; b < 8; ++b) acc[a][b] = {0.f, 0.f, 0.f, 0.f};
 
   const int l16 = lane & 15;              // fragment row/col
   const int kgrp = lane >> 4;             // 0..3 -> k-subblock of 8
-  const bool mvec = (m % 8 == 0);         // 16-B global loads legal
+  const bool vec_ok = (cpitch % 8 == 0);  // 16-B aligned row starts
 
-  // staging assignment: thread -> (k-row, 8-col segment); 512 slots per
-  // operand pair = 2 iterations of 256 threads
-  for (int kb = kb0; kb < kb1; ++kb) {
-    const int krow0 = kb * SY_BK;
+  // staging: thread -> (column, 8-deep k segment); 256 cols x 4 segs =
+  // 1024 slots per operand = 2 iterations of 512 threads.  Adjacent 4
+  // lanes take the 4 k-segments of one column (64-B half-lines; the other
+  // half of each 128-B line is the next k-block's data and hits L2).
+  //
+  // Single-buffer T14 register pipeline: the loads for block kb+1 are
+  // issued before the MFMA phase of block kb, which covers their latency;
+  // the per-CU VMEM return rate (~10 B/cyc HBM-bound) is the wall, which
+  // is why the tile is as large as LDS allows (bytes/flop ~ (M+N)/(M*N)).
+  auto load4 = [&](const bf16* mat, int c0, int col, int gk0) -> uint4 {
+    uint4 v;
+    bf16* vals = (bf16*)&v;
+    const int gc = c0 + col;
+    if (gc < m && vec_ok && gk0 + 8 <= c) {
+      v = *(const uint4*)(mat + (size_t)gc * cpitch + gk0);
+    } else if (gc < m) {
 #pragma unroll
-    for (int half = 0; half < 2; ++half) {
-      const int f = tid + half * 256;
-      const int kk = f >> 4;              // 0..31
-      const int seg = f & 15;
-      const int col = seg * 8;
-      const int gk = krow0 + kk;
-      auto stage = [&](const bf16* mat, int c0, bf16* dst) {
-        bf16 vals[8];
-        if (gk < c && mvec && c0 + col + 8 <= m) {
-          *(uint4*)vals = *(const uint4*)(mat + (size_t)gk * m + c0 + col);
-        } else if (gk < c) {
-          const bf16* src = mat + (size_t)gk * m;
+      for (int u = 0; u < 8; ++u)
+        vals[u] = (gk0 + u < c) ? mat[(size_t)gc * cpitch + gk0 + u]
+                                : (bf16)0.f;
+    } else {
 #pragma unroll
-          for (int u = 0; u < 8; ++u) {
-            const int gc = c0 + col + u;
-            vals[u] = (gc < m) ? src[gc] : (bf16)0.f;
-          }
-        } else {
+      for (int u = 0; u < 8; ++u) vals[u] = (bf16)0.f;
+    }
+    return v;
+  };
+
+  uint4 sv[2][4];                         // [iteration][operand] prefetch
+  auto load_all = [&](int kb) {
 #pragma unroll
-          for (int u = 0; u < 8; ++u) vals[u] = (bf16)0.f;
-        }
-        *(uint4*)&dst[kk * SY_STR + col] = *(uint4*)vals;
-      };
-      stage(Kc, i0, lt);
-      stage(Kc, j0, rt);
+    for (int it = 0; it < 2; ++it) {
+      const int f = tid + it * SY_WG;     // 0..1023
+      const int col = f >> 2;
+      const int gk0 = kb * SY_BK + (f & 3) * 8;
+      sv[it][0] = load4(KcT, i0, col, gk0);
+      sv[it][1] = load4(KcT, j0, col, gk0);
       if (HILO) {
-        stage(Kl, i0, ltl);
-        stage(Kl, j0, rtl);
+        sv[it][2] = load4(KlT, i0, col, gk0);
+        sv[it][3] = load4(KlT, j0, col, gk0);
       }
     }
-    __syncthreads();
-
-    {
+  };
+  auto write_all = [&]() {
 #pragma unroll
-      for (int a = 0; a < 4; ++a) {
-        const int arow = wr + a * 16 + l16;
-        bf16x8 afrag, afl;
-#pragma unroll
-        for (int t = 0; t < 8; ++t)
-          afrag[t] = lt[(kgrp * 8 + t) * SY_STR + arow];
-        if (HILO) {
-#pragma unroll
-          for (int t = 0; t < 8; ++t)
-            afl[t] = ltl[(kgrp * 8 + t) * SY_STR + arow];
-        }
-#pragma unroll
-        for (int b = 0; b < 4; ++b) {
-          const int bcol = wc + b * 16 + l16;
-          bf16x8 bfrag;
-#pragma unroll
-          for (int t = 0; t < 8; ++t)
-            bfrag[t] = rt[(kgrp * 8 + t) * SY_STR + bcol];
-          acc[a][b] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              afrag, bfrag, acc[a][b], 0, 0, 0);
-          if (HILO) {
-            bf16x8 bfl;
-#pragma unroll
-            for (int t = 0; t < 8; ++t)
-              bfl[t] = rtl[(kgrp * 8 + t) * SY_STR + bcol];
-            acc[a][b] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                afrag, bfl, acc[a][b], 0, 0, 0);
-            acc[a][b] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                afl, bfrag, acc[a][b], 0, 0, 0);
-          }
-        }
+    for (int it = 0; it < 2; ++it) {
+      const int f = tid + it * SY_WG;
+      const int o = (f >> 2) * SY_STR + (f & 3) * 8;
+      *(uint4*)&lt[o] = sv[it][0];
+      *(uint4*)&rt[o] = sv[it][1];
+      if (HILO) {
+        *(uint4*)&ltl[o] = sv[it][2];
+        *(uint4*)&rtl[o] = sv[it][3];
       }
+    }
+  };
+  auto mfma_pass = [&](const bf16* at, const bf16* bt) {
+    // one product pass in two b-halves of 16 INDEPENDENT MFMAs (no
+    // accumulator chaining between consecutive issues); fb covers 4 of
+    // the 8 column tiles at a time to stay inside the register budget
+    bf16x8 fa[4], fb[4];
+    const int ko = kgrp * 8;
+#pragma unroll
+    for (int a = 0; a < 4; ++a)
+      fa[a] = *(const bf16x8*)&at[(wr + a * 16 + l16) * SY_STR + ko];
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+#pragma unroll
+      for (int b = 0; b < 4; ++b)
+        fb[b] = *(const bf16x8*)&bt[(wc + (h * 4 + b) * 16 + l16) * SY_STR
+                                    + ko];
+#pragma unroll
+      for (int a = 0; a < 4; ++a)
+#pragma unroll
+        for (int b = 0; b < 4; ++b)
+          acc[a][h * 4 + b] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              fa[a], fb[b], acc[a][h * 4 + b], 0, 0, 0);
+    }
+  };
+
+  load_all(kb0);
+  for (int kb = kb0; kb < kb1; ++kb) {
+    write_all();                          // LDS free: readers passed barrier
+    __syncthreads();
+    if (kb + 1 < kb1) load_all(kb + 1);   // covered by the MFMA phase
+    mfma_pass(lt, rt);                    // hi * hi
+    if (HILO) {
+      mfma_pass(lt, rtl);                 // hi * lo
+      mfma_pass(ltl, rt);                 // lo * hi
     }
     __syncthreads();
   }
@@ -246,7 +309,7 @@ syrk_bf16_kernel(const bf16* __restrict__ Kc,   // [c, m] hi part
 #pragma unroll
   for (int a = 0; a < 4; ++a)
 #pragma unroll
-    for (int b = 0; b < 4; ++b)
+    for (int b = 0; b < 8; ++b)
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int gi = i0 + wr + a * 16 + crow_base + r;
@@ -259,11 +322,10 @@ syrk_bf16_kernel(const bf16* __restrict__ Kc,   // [c, m] hi part
       }
 }
 
-// Wait: for diagonal tiles (ti == tj) the full 128x128 tile is computed and
-// written once (it contains both halves), so no mirror is needed there; the
-// mirror above fills the strict lower triangle from the strict upper tiles.
-// Elements of diagonal tiles below the diagonal are produced by the same
-// block (a/b loops cover the whole tile), so KK ends up fully populated.
+// Diagonal tiles (ti == tj) compute the full 128x128 tile and write it once
+// (the a/b loops cover both halves), so no mirror is needed there; the
+// mirror above fills the strict lower triangle from the strict upper tiles,
+// leaving KK fully populated.
 
 // ---------------------------------------------------------------------------
 // colsum_gemv: Ky[m] += Kc^T y (fp64 accumulate per block, one atomic per
@@ -294,29 +356,31 @@ colsum_gemv_kernel(const bf16* __restrict__ Kc,   // [c, m]
 // ---------------------------------------------------------------------------
 extern "C" hipError_t launch_cross_kernel_tile(
     const float* X, const float* A, const float* s2v, float amp,
-    int c, int m, int d, void* out, void* out_lo, int out_is_bf16,
-    hipStream_t stream) {
+    int c, int m, int d, void* out, void* out_lo, void* out_t,
+    void* out_lo_t, int out_is_bf16, hipStream_t stream) {
   dim3 grid((c + CK_TILE - 1) / CK_TILE, (m + CK_TILE - 1) / CK_TILE);
   hipLaunchKernelGGL(cross_kernel_tile_kernel, grid, dim3(256), 0, stream,
                      X, A, s2v, amp, c, m, d,
                      out_is_bf16 ? (bf16*)out : nullptr,
                      out_is_bf16 ? (bf16*)out_lo : nullptr,
+                     (bf16*)out_t, (bf16*)out_lo_t,
                      out_is_bf16 ? nullptr : (float*)out);
   return hipGetLastError();
 }
 
-extern "C" hipError_t launch_syrk_bf16(const void* Kc, const void* Kl,
-                                       int c, int m, int split_k, float* KK,
-                                       hipStream_t stream) {
-  const int ntile = (m + 127) / 128;
+extern "C" hipError_t launch_syrk_bf16(const void* KcT, const void* KlT,
+                                       int c, int m, int cpitch, int split_k,
+                                       float* KK, hipStream_t stream) {
+  const int ntile = (m + SY_CT - 1) / SY_CT;
   dim3 grid(ntile * ntile, split_k);
-  if (Kl) {
-    hipLaunchKernelGGL((syrk_bf16_kernel<true>), grid, dim3(256), 0, stream,
-                       (const bf16*)Kc, (const bf16*)Kl, c, m, ntile,
-                       split_k, KK);
+  if (KlT) {
+    hipLaunchKernelGGL((syrk_bf16_kernel<true>), grid, dim3(SY_WG), 0,
+                       stream, (const bf16*)KcT, (const bf16*)KlT, c, m,
+                       cpitch, ntile, split_k, KK);
   } else {
-    hipLaunchKernelGGL((syrk_bf16_kernel<false>), grid, dim3(256), 0, stream,
-                       (const bf16*)Kc, nullptr, c, m, ntile, split_k, KK);
+    hipLaunchKernelGGL((syrk_bf16_kernel<false>), grid, dim3(SY_WG), 0,
+                       stream, (const bf16*)KcT, nullptr, c, m, cpitch,
+                       ntile, split_k, KK);
   }
   return hipGetLastError();
 }

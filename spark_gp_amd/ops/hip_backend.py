@@ -169,14 +169,19 @@ def kmn_knm_and_kmny(kernel: Kernel, active: torch.Tensor,
     # MFMA SYRK keeps input-quantization error at fp32 class while running
     # at bf16 matrix-core rate
     KK = torch.zeros(m, m, dtype=torch.float32, device=X.device)
-    ntile = (m + 127) // 128
+    ntile = (m + 255) // 256
     tiles = ntile * (ntile + 1) // 2
-    split_k = max(1, min(64, (512 + tiles - 1) // tiles))
+    # fill the machine: the 512-thread/80-KB-LDS SYRK runs 1 block/CU ->
+    # 256 slots; more slices than that only adds atomic traffic
+    split_k = max(1, min(64, 256 // tiles))
     for s in range(0, n, chunk_rows):
         e = min(n, s + chunk_rows)
-        Kc, Kl = ext.cross_kernel_tile(X[s:e].contiguous(), act32, s2,
-                                       float(C), True, True)
-        ext.syrk_bf16_acc(Kc, Kl, KK, split_k)
+        # the cross kernel also emits transposed (k-contiguous) copies so
+        # the MFMA SYRK stages with full-line loads and b128 LDS traffic
+        Kc, Kl, KcT, KlT = ext.cross_kernel_tile(X[s:e].contiguous(), act32,
+                                                 s2, float(C), True, True,
+                                                 True)
+        ext.syrk_bf16_acc(KcT, KlT, KK, split_k)
         yc = y32[s:e].contiguous()
         ext.colsum_gemv_acc(Kc, yc, Ky)
         ext.colsum_gemv_acc(Kl, yc, Ky)

@@ -117,19 +117,31 @@ def test_cross_kernel_tile_vs_torch(dev, ext):
                                rtol=1e-4, atol=1e-5)
 
 
+def test_cross_kernel_transposed_outputs(dev, ext):
+    """want_t=True must return exact transposes of the hi/lo blocks."""
+    g = torch.Generator().manual_seed(9)
+    X = torch.rand(777, 16, generator=g).to(dev)    # c not multiple of 8
+    A = torch.rand(333, 16, generator=g).to(dev)
+    s2 = torch.full((16,), 0.7, device=dev)
+    hi, lo, hiT, loT = ext.cross_kernel_tile(X, A, s2, 1.3, True, True, True)
+    assert torch.equal(hiT, hi.T.contiguous())
+    assert torch.equal(loT, lo.T.contiguous())
+
+
 def test_syrk_bf16_vs_matmul(dev, ext):
     g = torch.Generator().manual_seed(5)
     # asymmetric, non-tile-multiple shapes to catch transposes and guards
     c, m = 1000, 333
     Kc = (torch.rand(c, m, generator=g) * 2 - 0.5).to(dev).bfloat16()
+    KcT = Kc.T.contiguous()          # SYRK takes the transposed block [m, c]
     KK = torch.zeros(m, m, dtype=torch.float32, device=dev)
-    ext.syrk_bf16_acc(Kc, None, KK, 4)
+    ext.syrk_bf16_acc(KcT, None, KK, 4)
     ref = (Kc.float().T @ Kc.float())
     diff = (KK - ref).abs()
     denom = ref.abs().clamp_min(1.0)
     assert float((diff / denom).max()) < 2e-2
     # accumulation semantics: second call doubles
-    ext.syrk_bf16_acc(Kc, None, KK, 4)
+    ext.syrk_bf16_acc(KcT, None, KK, 4)
     assert float(((KK - 2 * ref).abs() / denom.clamp_min(2.0)).max()) < 3e-2
 
 
@@ -141,7 +153,7 @@ def test_syrk_hilo_split_accuracy(dev, ext):
     hi = V.bfloat16()
     lo = (V - hi.float()).bfloat16()
     KK = torch.zeros(m, m, dtype=torch.float32, device=dev)
-    ext.syrk_bf16_acc(hi, lo, KK, 8)
+    ext.syrk_bf16_acc(hi.T.contiguous(), lo.T.contiguous(), KK, 8)
     ref = (V.double().T @ V.double())
     rel = float(((KK.double() - ref).abs() / ref.abs().clamp_min(1.0)).max())
     # bf16-only would be ~1e-3 here; hi/lo must be well under 1e-4
