@@ -146,7 +146,7 @@ cross_kernel_tile_kernel(const float* __restrict__ X,    // [c, d]
 }
 
 // ---------------------------------------------------------------------------
-// syrk_bf16: KK[m, m] += Kc^T Kc, Kc [c, m] bf16 row-major
+// syrk_bf16: KK[m, m] += K^T K from the TRANSPOSED chunk KcT [m, c]
 // ---------------------------------------------------------------------------
 // 512-thread block = 8 waves in a 4x2 grid; 256x256 output tile per block
 // (the largest tile the LDS+register budget allows: per-CU VMEM return
@@ -322,7 +322,7 @@ syrk_bf16_kernel(const bf16* __restrict__ KcT,  // [m, cpitch] hi^T
       }
 }
 
-// Diagonal tiles (ti == tj) compute the full 128x128 tile and write it once
+// Diagonal tiles (ti == tj) compute the full SY_CT x SY_CT tile and write it
 // (the a/b loops cover both halves), so no mirror is needed there; the
 // mirror above fills the strict lower triangle from the strict upper tiles,
 // leaving KK fully populated.

@@ -111,3 +111,27 @@ def test_group_experts_partition():
         k = Xg.shape[1]
         mat = idx.reshape(Xg.shape[0], k)
         assert ((mat % E) == mat[:, :1] % E).all()
+
+
+def test_predictive_variance_shrinks_at_training_points():
+    """PPA predictive variance must be ~sigma2-level at training inputs and
+    grow far from the data (Rasmussen & Williams eq. 8.27 behavior)."""
+    import torch
+    from spark_gp_amd import GaussianProcessRegression
+    from spark_gp_amd.kernels import ARDRBFKernel
+
+    rng = np.random.default_rng(5)
+    X = rng.uniform(size=(600, 2))
+    y = np.sin(4 * X.sum(-1)) + 0.05 * rng.normal(size=600)
+    model = (GaussianProcessRegression()
+             .setKernel(lambda: 1 * ARDRBFKernel(2))
+             .setDatasetSizeForExpert(60).setActiveSetSize(200)
+             .setSigma2(1e-2).setMaxIter(30).setSeed(0).setDevice("cpu")
+             .fit(X, y))
+    _, std_in = model.predict(X[:50], return_std=True)
+    far = np.full((10, 2), 25.0)          # far outside the unit square
+    mean_far, std_far = model.predict(far, return_std=True)
+    assert std_in.mean() < 0.35
+    assert std_far.min() > 3 * std_in.mean()
+    # far from data the PPA mean falls back toward the prior mean 0
+    assert np.abs(mean_far).max() < 0.1
