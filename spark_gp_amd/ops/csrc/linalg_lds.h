@@ -110,66 +110,83 @@ __device__ inline void chol_invert_lower(float* Abuf, float* Tbuf,
       for (int q = 0; q < nq; ++q) {
         const int qb = q * 8;
         const int sbs = min(8, bs - qb);
-        if (lane == 0) {
-          bool nonfin = false;
-          float m[8][8];
+        {
+          // 8x8 factor + trtri on lanes 0..7, column j per lane, cross-lane
+          // traffic via __shfl (ds_bpermute).  Replaces a lane-0 register
+          // version whose ~1200 single-lane instructions dominated the
+          // kernel (a lane-parallel in-LDS variant measured 20% worse
+          // still — LDS round-trip latency; see git history).  Working set
+          // is ~32 registers, so no VGPR pressure.  Lanes 8..63 compute
+          // duplicates and are masked off every store.
+          const int j = lane & 7;
+          float col[8];                 // this lane's column: col[i]=m[i][j]
 #pragma unroll
           for (int i = 0; i < 8; ++i)
-#pragma unroll
-            for (int c = 0; c < 8; ++c)
-              m[i][c] = (i < sbs && c <= i)
-                            ? D[(size_t)(qb + i) * SA + qb + c]
-                            : (i == c ? 1.f : 0.f);
+            col[i] = (j < sbs && i < sbs && i >= j)
+                         ? D[(size_t)(qb + i) * SA + qb + j]
+                         : (i == j ? 1.f : 0.f);
+          bool ok = true, nonfin = false;
           double ldet = 0.0;
-          bool ok = true;
 #pragma unroll
           for (int ss = 0; ss < 8; ++ss) {
-            const float piv = m[ss][ss];
-            if (ss < sbs && ok) {     // classify only the FIRST failure
+            float l[8];                 // owner lane ss broadcasts column ss
+#pragma unroll
+            for (int i = 0; i < 8; ++i) l[i] = __shfl(col[i], ss, 64);
+            const float piv = l[ss];
+            if (ss < sbs && ok) {       // classify only the FIRST failure
               if (!isfinite(piv)) { ok = false; nonfin = true; }
               else if (!(piv > 0.f)) ok = false;
               else ldet += (double)__logf(piv);
             }
             const float rs = rsqrtf(piv);
-            m[ss][ss] = piv * rs;
 #pragma unroll
-            for (int i = ss + 1; i < 8; ++i) m[i][ss] *= rs;
+            for (int i = 0; i < 8; ++i) l[i] *= rs;      // l[ss] = piv*rs
+            if (j == ss) {
 #pragma unroll
-            for (int i = ss + 1; i < 8; ++i)
+              for (int i = 0; i < 8; ++i)
+                if (i >= ss) col[i] = l[i];              // keep scaled column
+            } else if (j > ss) {
+              const float ljs = l[j];
 #pragma unroll
-              for (int c = ss + 1; c <= i; ++c)
-                m[i][c] -= m[i][ss] * m[c][ss];
+              for (int i = 0; i < 8; ++i)
+                if (i >= j) col[i] -= l[i] * ljs;        // rank-1 update
+            }
           }
-          // write L back; in-register trtri8 into Vq
+          if (lane < 8 && j < sbs) {    // write L back (column j)
 #pragma unroll
-          for (int i = 0; i < 8; ++i)
+            for (int i = 0; i < 8; ++i)
+              if (i >= j && i < sbs)
+                D[(size_t)(qb + i) * SA + qb + j] = col[i];
+          }
+          // trtri8: V column j by forward substitution; row i of L is
+          // gathered as rowi[c] = shfl(col[i], c)
+          float v[8];
 #pragma unroll
-            for (int c = 0; c <= i; ++c)
-              if (i < sbs) D[(size_t)(qb + i) * SA + qb + c] = m[i][c];
-          float v8[8][8];
+          for (int i = 0; i < 8; ++i) v[i] = 0.f;
+          v[j] = __builtin_amdgcn_rcpf(col[j]);
 #pragma unroll
           for (int i = 0; i < 8; ++i) {
-            const float rli = __builtin_amdgcn_rcpf(m[i][i]);
+            float rowi[8];
 #pragma unroll
-            for (int j = 0; j < 8; ++j) {
-              if (j > i) { v8[i][j] = 0.f; continue; }
-              if (j == i) { v8[i][j] = rli; continue; }
+            for (int c = 0; c < 8; ++c) rowi[c] = __shfl(col[i], c, 64);
+            if (j < i) {
               float sacc = 0.f;
 #pragma unroll
               for (int c = 0; c < 8; ++c)
-                if (c >= j && c < i) sacc += m[i][c] * v8[c][j];
-              v8[i][j] = -sacc * rli;
+                if (c >= j && c < i) sacc += rowi[c] * v[c];
+              v[i] = -sacc * __builtin_amdgcn_rcpf(rowi[i]);
             }
           }
+          if (lane < 8) {
 #pragma unroll
-          for (int i = 0; i < 8; ++i)
-#pragma unroll
-            for (int j = 0; j < 8; ++j)
-              Vq[q * 64 + i * 8 + j] = v8[i][j];
-          if (!ok) {
-            if (*bad == 0) *bad = nonfin ? 2 : 1;  // sticky: first cause
-          } else {
-            misc[0] += ldet;
+            for (int i = 0; i < 8; ++i) Vq[q * 64 + i * 8 + j] = v[i];
+          }
+          if (lane == 0) {
+            if (!ok) {
+              if (*bad == 0) *bad = nonfin ? 2 : 1;  // sticky: first cause
+            } else {
+              misc[0] += ldet;
+            }
           }
         }
         __syncwarp();
