@@ -42,8 +42,12 @@ total = mean.sum()
 print(f"sum of phases: {total:.2f} us (mean per expert)")
 for i in range(9):
     print(f"  {phases[i]:>8}: {mean[i]:8.2f} us  ({100*mean[i]/total:4.1f}%)")
-c1 = clk[:, 10].mean() / FREQ * 1e6
-print(f"C1 (diag factor+inv) within C: {c1:.2f} us")
+sub = clk[:, 12:18].mean(0) / FREQ * 1e6
+subnames = ["phase1 diag-pre", "phase2 wall", "wave0 span",
+            "waves1-7 span", "C2 panel solve", "D off-diag trtri"]
+print("C/D sub-phases (accumulated over J, mean per expert):")
+for n, v in zip(subnames, sub):
+    print(f"  {n:>18}: {v:8.2f} us")
 # wall span of the whole launch from clocks
 span = (clk[:, 9].max() - clk[:, 0].min()) / FREQ * 1e3
 print(f"launch span by clocks: {span:.2f} ms")

@@ -69,7 +69,7 @@ std::vector<torch::Tensor> fused_expert_nll_impl(torch::Tensor X,
   torch::Tensor clk;
   unsigned long long* clk_ptr = nullptr;
   if (profile) {
-    clk = torch::zeros({E, 12},
+    clk = torch::zeros({E, 20},
                        torch::TensorOptions().dtype(torch::kInt64)
                            .device(X.device()));
     clk_ptr = (unsigned long long*)clk.data_ptr<int64_t>();

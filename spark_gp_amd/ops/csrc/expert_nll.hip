@@ -102,11 +102,11 @@ fused_expert_nll_kernel(const float* __restrict__ Xg,
                         double* __restrict__ out_trG,      // [E]
                         double* __restrict__ out_contr,    // [E, d]
                         int* __restrict__ out_bad,
-                        unsigned long long* __restrict__ out_clk) { // [E,12]
+                        unsigned long long* __restrict__ out_clk) { // [E,20]
                         // optional phase profiling (wall_clock64 boundaries)
   extern __shared__ char lds_raw[];
 #define PH(n) do { if (out_clk && threadIdx.x == 0) \
-    out_clk[(size_t)blockIdx.x * 12 + (n)] = wall_clock64(); } while (0)
+    out_clk[(size_t)blockIdx.x * 20 + (n)] = wall_clock64(); } while (0)
   PH(0);
   NllLds S = carve(lds_raw, k, d);
   const int SA = k + 1;
@@ -160,7 +160,8 @@ PH(2);
   // ---- C/D: in-place blocked Cholesky + triangular inverse ---------
   // (shared machinery: linalg_lds.h)  A: lower K -> V = L^-1; upper Kb
   // cache untouched; log|K| into misc[0]; bad flag on fp32 breakdown.
-  chol_invert_lower(S.A, S.T, k, SA, tid, lane, S.bad, S.misc);
+  chol_invert_lower(S.A, S.T, k, SA, tid, lane, S.bad, S.misc,
+                    out_clk ? out_clk + (size_t)e * 20 + 12 : nullptr);
   if (*S.bad) {
     if (tid == 0) {
       out_bad[e] = *S.bad;     // 1: indefinite, 2: non-finite iterate

@@ -65,7 +65,11 @@ __device__ inline float dot4(const float* p, int sp, const float* q, int sq,
 __device__ inline void chol_invert_lower(float* Abuf, float* Tbuf,
                                          const int k, const int SA,
                                          const int tid, const int lane,
-                                         int* bad, double* misc) {
+                                         int* bad, double* misc,
+                                         unsigned long long* jclk = nullptr) {
+  // jclk (profiling only): accumulated cycles into
+  //   [0] phase1  [1] phase2 wall  [2] wave0 span  [3] waves1-7 span
+  //   [4] C2 panel solve  [5] D off-diag trtri
   const int nblk = (k + NB - 1) / NB;
     // ---- C: blocked in-place Cholesky with look-ahead ----------------
   // Right-looking, restructured so the serial diagonal factorization
@@ -79,6 +83,8 @@ __device__ inline void chol_invert_lower(float* Abuf, float* Tbuf,
     const int bs = min(NB, k - jb);
     float* D = Abuf + (size_t)jb * SA + jb;   // diag block, stride SA
     const int pj = jb - NB;                  // previous panel column offset
+    unsigned long long jt0 = 0, jt1 = 0;
+    if (jclk && tid == 0) jt0 = wall_clock64();
 
     if (J > 0) {
       // phase1: update ONLY the diagonal block (rows/cols jb..jb+bs, c<=i)
@@ -94,22 +100,26 @@ __device__ inline void chol_invert_lower(float* Abuf, float* Tbuf,
       }
       __syncthreads();
     }
+    if (jclk && (tid == 0 || tid == 64)) jt1 = wall_clock64();
 
-    if (tid < 64) {
-      // wave 0: factor + invert the bs x bs diagonal block via 8x8
-      // sub-blocks.  Each 8x8 sub-diagonal is Cholesky-factored AND
-      // inverted entirely inside lane 0's registers (fully unrolled, the
-      // serial dependency chain never touches LDS); panels and trailing
-      // updates are lane-parallel; the 32-level inverse assembles from the
-      // 8x8 inverses by block back-substitution.  8x8 inverses live in
-      // T[0..256); T[256..448) is shell scratch.
-      const int row = lane & 31;
-      const int nq = (bs + 7) / 8;
-      float* Vq = Tbuf;               // [nq][8][8]
-      float* TS = Tbuf + 256;         // shell scratch [3][8][8]
-      for (int q = 0; q < nq; ++q) {
-        const int qb = q * 8;
-        const int sbs = min(8, bs - qb);
+    // phase2, restructured (round-1 profiling: the old wave-0-only
+    // version spent 86 us/expert on one wave while waves 1-7 idled 5 us):
+    // per 8-column sub-block q, wave 0 runs the shfl factor+trtri while
+    // waves 1-7 process chunk q of panel J-1's update to the rows below
+    // this column block; then ALL 512 threads apply the intra-block
+    // panel/trailing for q.  The 32x32 inverse assembly afterwards also
+    // runs on all threads.  8x8 inverses live in T[0..256); T[256..448)
+    // is shell scratch.
+    const int nq = (bs + 7) / 8;
+    float* Vq = Tbuf;               // [nq][8][8]
+    float* TS = Tbuf + 256;         // shell scratch [3][8][8]
+    const int t0r = jb + bs;        // first row below this column block
+    const int nrp = k - t0r;
+    const int ntri_rest = nrp * (nrp + 1) / 2;
+    for (int q = 0; q < nq; ++q) {
+      const int qb = q * 8;
+      const int sbs = min(8, bs - qb);
+      if (tid < 64) {
         {
           // 8x8 factor + trtri on lanes 0..7, column j per lane, cross-lane
           // traffic via __shfl (ds_bpermute).  Replaces a lane-0 register
@@ -126,17 +136,27 @@ __device__ inline void chol_invert_lower(float* Abuf, float* Tbuf,
                          ? D[(size_t)(qb + i) * SA + qb + j]
                          : (i == j ? 1.f : 0.f);
           bool ok = true, nonfin = false;
-          double ldet = 0.0;
+          // log|sub-block| via one logf at the end: accumulate the pivot
+          // mantissa product and exponent sum (branchless flag updates —
+          // the straightforward per-step `ldet += (double)__logf(piv)`
+          // plus isfinite/positivity branches cost ~35 instructions and 3
+          // exec-mask branches PER PIVOT in the generated ISA)
+          float mprod = 1.f;
+          int expsum = 0;
 #pragma unroll
           for (int ss = 0; ss < 8; ++ss) {
             float l[8];                 // owner lane ss broadcasts column ss
 #pragma unroll
             for (int i = 0; i < 8; ++i) l[i] = __shfl(col[i], ss, 64);
             const float piv = l[ss];
-            if (ss < sbs && ok) {       // classify only the FIRST failure
-              if (!isfinite(piv)) { ok = false; nonfin = true; }
-              else if (!(piv > 0.f)) ok = false;
-              else ldet += (double)__logf(piv);
+            if (ss < sbs) {
+              const bool fin = isfinite(piv);
+              // classify only the FIRST failure (nonfin wins only if it
+              // happens while still ok)
+              nonfin = nonfin || (ok && !fin);
+              ok = ok && fin && (piv > 0.f);
+              mprod *= __builtin_amdgcn_frexp_mantf(piv);
+              expsum += __builtin_amdgcn_frexp_expf(piv);
             }
             const float rs = rsqrtf(piv);
 #pragma unroll
@@ -185,53 +205,81 @@ __device__ inline void chol_invert_lower(float* Abuf, float* Tbuf,
             if (!ok) {
               if (*bad == 0) *bad = nonfin ? 2 : 1;  // sticky: first cause
             } else {
-              misc[0] += ldet;
+              misc[0] += (double)__logf(mprod)
+                         + (double)expsum * 0.6931471805599453;
             }
           }
         }
-        __syncwarp();
-        if (*bad) break;               // wave-uniform: skip garbage blocks
-        // panel rows within the 32-block: P = A * Vq^T; then trailing
-        const int p0 = qb + sbs;       // first panel row (local)
-        const int pr = bs - p0;        // panel rows
-        if (pr > 0) {
-          // panel: (pr x sbs) elements, lanes parallel
-          for (int f = lane; f < pr * 8; f += 64) {
-            const int r = f >> 3, c = f & 7;
-            if (c >= sbs) continue;
-            float sacc = 0.f;
-            const float* ar = D + (size_t)(p0 + r) * SA + qb;
-            const float* vr = Vq + q * 64 + c * 8;
-            for (int t = 0; t <= c; ++t) sacc += ar[t] * vr[t];
-            TS[f] = sacc;              // hold row piece until all reads done
-          }
-          __syncwarp();
-          for (int f = lane; f < pr * 8; f += 64) {
-            const int r = f >> 3, c = f & 7;
-            if (c >= sbs) continue;
-            D[(size_t)(p0 + r) * SA + qb + c] = TS[f];
-          }
-          __syncwarp();
-          // trailing: lower incl diag of remaining rows
-          const int ntri = pr * (pr + 1) / 2;
-          for (int f = lane; f < ntri; f += 64) {
-            int a, b;
-            tri_decode(f, a, b);
-            const int i = p0 + a, c = p0 + b;
-            Abuf[(size_t)(jb + i) * SA + jb + c] -=
-                dot4(D + (size_t)i * SA + qb, 1, D + (size_t)c * SA + qb, 1,
-                     0, sbs);
-          }
-          __syncwarp();
+      } else if (J > 0) {
+        // waves 1-7, chunk q (strided by nq, so any nq covers everything):
+        // previous panel's update to (a) the panel rows of column-block J
+        // and (b) the remaining trailing triangle
+        for (int f = (tid - 64) + q * (WG - 64); f < nrp * bs;
+             f += (WG - 64) * nq) {
+          const int r = f / bs, c = f - r * bs;
+          const int i = t0r + r, cc = jb + c;
+          Abuf[(size_t)i * SA + cc] -=
+              dot4(Abuf + (size_t)i * SA + pj, 1,
+                   Abuf + (size_t)cc * SA + pj, 1, 0, NB);
+        }
+        for (int f = (tid - 64) + q * (WG - 64); f < ntri_rest;
+             f += (WG - 64) * nq) {
+          int a, b;
+          tri_decode(f, a, b);
+          const int i = t0r + a, c = t0r + b;
+          Abuf[(size_t)i * SA + c] -=
+              dot4(Abuf + (size_t)i * SA + pj, 1,
+                   Abuf + (size_t)c * SA + pj, 1, 0, NB);
         }
       }
+      __syncthreads();
+      if (*bad) break;               // uniform: read after the barrier
+      // intra-block panel P = A * Vq^T and trailing for q, ALL threads
+      const int p0 = qb + sbs;       // first panel row (local)
+      const int pr = bs - p0;        // panel rows
+      if (pr > 0) {
+        for (int f = tid; f < pr * 8; f += WG) {
+          const int r = f >> 3, c = f & 7;
+          if (c >= sbs) continue;
+          float sacc = 0.f;
+          const float* ar = D + (size_t)(p0 + r) * SA + qb;
+          const float* vr = Vq + q * 64 + c * 8;
+          for (int t = 0; t <= c; ++t) sacc += ar[t] * vr[t];
+          TS[f] = sacc;              // hold row piece until all reads done
+        }
+        __syncthreads();
+        for (int f = tid; f < pr * 8; f += WG) {
+          const int r = f >> 3, c = f & 7;
+          if (c >= sbs) continue;
+          D[(size_t)(p0 + r) * SA + qb + c] = TS[f];
+        }
+        __syncthreads();
+        // trailing: lower incl diag of remaining rows of this block
+        const int ntri = pr * (pr + 1) / 2;
+        for (int f = tid; f < ntri; f += WG) {
+          int a, b;
+          tri_decode(f, a, b);
+          const int i = p0 + a, c = p0 + b;
+          Abuf[(size_t)(jb + i) * SA + jb + c] -=
+              dot4(D + (size_t)i * SA + qb, 1, D + (size_t)c * SA + qb, 1,
+                   0, sbs);
+        }
+        __syncthreads();
+      }
+    }
+    unsigned long long at0 = 0;
+    if (jclk && tid == 0) {
+      at0 = wall_clock64();
+      jclk[2] += at0 - jt1;          // q-loop wall (factor+chunks+intra)
+    }
+    if (*bad == 0) {
       // ---- assemble V_JJ (32x32 inverse) from the 8x8 inverses --------
       // column-blocks Jq descending; all target blocks of a column in
-      // parallel through TS
+      // parallel through TS; ALL threads
       for (int Jq = nq - 1; Jq >= 0; --Jq) {
         const int jb8 = Jq * 8;
         const int nblks = nq - 1 - Jq;       // target blocks below
-        for (int f = lane; f < nblks * 64; f += 64) {
+        for (int f = tid; f < nblks * 64; f += WG) {
           const int blk = f >> 6;            // 0..nblks-1
           const int ib = (Jq + 1 + blk) * 8;
           const int i = (f >> 3) & 7, j = f & 7;
@@ -244,8 +292,8 @@ __device__ inline void chol_invert_lower(float* Abuf, float* Tbuf,
           }
           TS[f] = u;
         }
-        __syncwarp();
-        for (int f = lane; f < nblks * 64; f += 64) {
+        __syncthreads();
+        for (int f = tid; f < nblks * 64; f += WG) {
           const int blk = f >> 6;
           const int ib = (Jq + 1 + blk) * 8;
           const int i = (f >> 3) & 7, j = f & 7;
@@ -260,37 +308,22 @@ __device__ inline void chol_invert_lower(float* Abuf, float* Tbuf,
         }
         // this column's diagonal block <- its inverse (consumed as V by
         // the shells of columns further left)
-        for (int f = lane; f < 64; f += 64) {
+        for (int f = tid; f < 64; f += WG) {
           const int i = (f >> 3) & 7, j = f & 7;
           const int gi = jb8 + i, gj = jb8 + j;
           if (gi < bs && gj <= gi) D[(size_t)gi * SA + gj] = Vq[Jq * 64 + f];
         }
-        __syncwarp();
-      }
-      __syncwarp();
-    } else if (J > 0) {
-      // waves 1-7: previous panel's update to (a) the panel rows of
-      // column-block J and (b) the remaining trailing triangle
-      const int t0r = jb + bs;
-      const int nrp = k - t0r;
-      for (int f = tid - 64; f < nrp * bs; f += WG - 64) {
-        const int r = f / bs, c = f - r * bs;
-        const int i = t0r + r, cc = jb + c;
-        Abuf[(size_t)i * SA + cc] -=
-            dot4(Abuf + (size_t)i * SA + pj, 1,
-                 Abuf + (size_t)cc * SA + pj, 1, 0, NB);
-      }
-      const int ntri = nrp * (nrp + 1) / 2;
-      for (int f = tid - 64; f < ntri; f += WG - 64) {
-        int a, b;
-        tri_decode(f, a, b);
-        const int i = t0r + a, c = t0r + b;
-        Abuf[(size_t)i * SA + c] -=
-            dot4(Abuf + (size_t)i * SA + pj, 1,
-                 Abuf + (size_t)c * SA + pj, 1, 0, NB);
+        __syncthreads();
       }
     }
+    if (jclk && tid == 0) jclk[3] += wall_clock64() - at0;  // assembly wall
     __syncthreads();
+    if (jclk && tid == 0) {
+      const unsigned long long t2 = wall_clock64();
+      jclk[0] += jt1 - jt0;
+      jclk[1] += t2 - jt1;
+      jt0 = t2;
+    }
     if (*bad) break;
 
     const int t0 = jb + bs;        // first trailing row
@@ -310,9 +343,12 @@ __device__ inline void chol_invert_lower(float* Abuf, float* Tbuf,
       }
       __syncthreads();
     }
+    if (jclk && tid == 0) jclk[4] += wall_clock64() - jt0;
   }
 
   if (*bad) return;
+  unsigned long long dt0 = 0;
+  if (jclk && tid == 0) dt0 = wall_clock64();
   // ---- D: off-diagonal triangular inverse, in place, J right-to-left
   // V_IJ = -(sum_{K=J+1..I} V_IK L_KJ) L_JJ^-1 ; rows fully parallel.
   for (int J = nblk - 2; J >= 0; --J) {
@@ -336,5 +372,5 @@ __device__ inline void chol_invert_lower(float* Abuf, float* Tbuf,
     }
     __syncthreads();
   }
-
+  if (jclk && tid == 0) jclk[5] += wall_clock64() - dt0;
 }
