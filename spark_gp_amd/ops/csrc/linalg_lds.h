@@ -121,33 +121,35 @@ __device__ inline void chol_invert_lower(float* Abuf, float* Tbuf,
       const int sbs = min(8, bs - qb);
       if (tid < 64) {
         {
-          // 8x8 factor + trtri on lanes 0..7, column j per lane, cross-lane
-          // traffic via __shfl (ds_bpermute).  Replaces a lane-0 register
-          // version whose ~1200 single-lane instructions dominated the
-          // kernel (a lane-parallel in-LDS variant measured 20% worse
-          // still — LDS round-trip latency; see git history).  Working set
-          // is ~32 registers, so no VGPR pressure.  Lanes 8..63 compute
-          // duplicates and are masked off every store.
+          // 8x8 factor + trtri on lanes 0..7, ROW j per lane, cross-lane
+          // traffic via __shfl (ds_bpermute).  Row-per-lane keeps every
+          // runtime-varying access at a STATIC register index: the rank-1
+          // scalar is the lane's own row[ss], and the update bounds
+          // c in (ss, 7] are compile-time — the earlier column-per-lane
+          // version needed l[j] with a runtime lane index, which compiles
+          // to an 8-deep v_cmp/cndmask select chain per pivot step.
+          // Lanes 8..63 compute duplicates and are masked off every store.
           const int j = lane & 7;
-          float col[8];                 // this lane's column: col[i]=m[i][j]
+          float row[8];                 // this lane's row: row[c] = m[j][c]
 #pragma unroll
-          for (int i = 0; i < 8; ++i)
-            col[i] = (j < sbs && i < sbs && i >= j)
-                         ? D[(size_t)(qb + i) * SA + qb + j]
-                         : (i == j ? 1.f : 0.f);
+          for (int c = 0; c < 8; ++c)
+            row[c] = (j < sbs && c < sbs && c <= j)
+                         ? D[(size_t)(qb + j) * SA + qb + c]
+                         : (c == j ? 1.f : 0.f);
           bool ok = true, nonfin = false;
           // log|sub-block| via one logf at the end: accumulate the pivot
-          // mantissa product and exponent sum (branchless flag updates —
-          // the straightforward per-step `ldet += (double)__logf(piv)`
-          // plus isfinite/positivity branches cost ~35 instructions and 3
-          // exec-mask branches PER PIVOT in the generated ISA)
+          // mantissa product and exponent sum (branchless — a per-step
+          // `ldet += (double)__logf(piv)` plus isfinite/positivity
+          // branches cost ~35 instructions per pivot)
           float mprod = 1.f;
           int expsum = 0;
+          float myrs = 1.f;             // this row's pivot 1/sqrt
+          float rsv = 1.f;
 #pragma unroll
           for (int ss = 0; ss < 8; ++ss) {
-            float l[8];                 // owner lane ss broadcasts column ss
+            float l[8];                 // column ss: l[c] = lane c's row[ss]
 #pragma unroll
-            for (int i = 0; i < 8; ++i) l[i] = __shfl(col[i], ss, 64);
+            for (int c = 0; c < 8; ++c) l[c] = __shfl(row[ss], c, 64);
             const float piv = l[ss];
             if (ss < sbs) {
               const bool fin = isfinite(piv);
@@ -159,36 +161,35 @@ __device__ inline void chol_invert_lower(float* Abuf, float* Tbuf,
               expsum += __builtin_amdgcn_frexp_expf(piv);
             }
             const float rs = rsqrtf(piv);
+            if (j == ss) myrs = rs;
+            // subtract l_j*l_c = (rs*row[ss]) * (rs*l[c]) for c > ss; rows
+            // j <= ss get w = 0 (their updates ended at step j)
+            const float w = (j > ss) ? rs * row[ss] : 0.f;
 #pragma unroll
-            for (int i = 0; i < 8; ++i) l[i] *= rs;      // l[ss] = piv*rs
-            if (j == ss) {
-#pragma unroll
-              for (int i = 0; i < 8; ++i)
-                if (i >= ss) col[i] = l[i];              // keep scaled column
-            } else if (j > ss) {
-              const float ljs = l[j];
-#pragma unroll
-              for (int i = 0; i < 8; ++i)
-                if (i >= j) col[i] -= l[i] * ljs;        // rank-1 update
-            }
+            for (int c = ss + 1; c < 8; ++c) row[c] -= (rs * l[c]) * w;
           }
-          if (lane < 8 && j < sbs) {    // write L back (column j)
+          // deferred column scaling: L[j][c] = row[c] * rs_c; gather the
+          // rs vector from the diagonal lanes (rs_c = lane c's myrs)
 #pragma unroll
-            for (int i = 0; i < 8; ++i)
-              if (i >= j && i < sbs)
-                D[(size_t)(qb + i) * SA + qb + j] = col[i];
+          for (int c = 0; c < 8; ++c) {
+            rsv = __shfl(myrs, c, 64);
+            row[c] *= rsv;
+          }
+          if (lane < 8 && j < sbs) {    // write L back (row j)
+#pragma unroll
+            for (int c = 0; c < 8; ++c)
+              if (c <= j) D[(size_t)(qb + j) * SA + qb + c] = row[c];
           }
           // trtri8: V column j by forward substitution; row i of L is
-          // gathered as rowi[c] = shfl(col[i], c)
+          // gathered as rowi[c] = lane i's row[c]
           float v[8];
 #pragma unroll
-          for (int i = 0; i < 8; ++i) v[i] = 0.f;
-          v[j] = __builtin_amdgcn_rcpf(col[j]);
+          for (int i = 0; i < 8; ++i) v[i] = (i == j) ? myrs : 0.f;
 #pragma unroll
           for (int i = 0; i < 8; ++i) {
             float rowi[8];
 #pragma unroll
-            for (int c = 0; c < 8; ++c) rowi[c] = __shfl(col[i], c, 64);
+            for (int c = 0; c < 8; ++c) rowi[c] = __shfl(row[c], i, 64);
             if (j < i) {
               float sacc = 0.f;
 #pragma unroll
