@@ -20,8 +20,9 @@ Quick start::
 
 from .active_set import (ActiveSetProvider, GreedilyOptimizingActiveSetProvider,
                          KMeansActiveSetProvider, RandomActiveSetProvider)
-from .kernels import (ARDRBFKernel, EyeKernel, Kernel, RBFKernel, Scalar,
-                      SumOfKernels, WhiteNoiseKernel)
+from .kernels import (ARDRBFKernel, EyeKernel, Kernel, Matern32Kernel,
+                      Matern52Kernel, RBFKernel, Scalar, SumOfKernels,
+                      WhiteNoiseKernel)
 from .models import (GaussianProcessClassificationModel,
                      GaussianProcessClassifier, GaussianProcessRegression,
                      GaussianProcessRegressionModel, load_model, save_model)
@@ -35,7 +36,8 @@ __version__ = "0.1.0"
 __all__ = [
     "GaussianProcessRegression", "GaussianProcessRegressionModel",
     "GaussianProcessClassifier", "GaussianProcessClassificationModel",
-    "Kernel", "RBFKernel", "ARDRBFKernel", "EyeKernel", "WhiteNoiseKernel",
+    "Kernel", "RBFKernel", "ARDRBFKernel", "Matern32Kernel",
+    "Matern52Kernel", "EyeKernel", "WhiteNoiseKernel",
     "SumOfKernels", "Scalar",
     "ActiveSetProvider", "RandomActiveSetProvider", "KMeansActiveSetProvider",
     "GreedilyOptimizingActiveSetProvider",
