@@ -92,12 +92,20 @@ std::vector<torch::Tensor> fused_expert_nll(torch::Tensor X, torch::Tensor y,
   return fused_expert_nll_impl(X, y, scale, amp, noise, false);
 }
 
+namespace {
+inline int64_t a16i(int64_t n) { return (n + 15) & ~(int64_t)15; }
+inline int64_t sa_of(int64_t k) { return (k + 4) & ~(int64_t)3; }
+inline int64_t tsz_of(int64_t k) {
+  return std::max<int64_t>(std::max<int64_t>(k * 36, 32 * sa_of(k)), 448);
+}
+}  // namespace
+
 bool fused_expert_nll_supported(int64_t k, int64_t d) {
   if (k > 128 || d > 128 || k < 1) return false;
-  // LDS budget: mirror of nll_lds_bytes2 in expert_nll.hip
-  const int64_t tsz = std::max<int64_t>(k * 33, 32 * (k + 1));
-  int64_t bytes = 8 * 10 + 4 * (k * (k + 1) + tsz + k * (d + 1) + 4 * k + d)
-                  + 16;
+  // LDS budget: EXACT mirror of nll_lds_bytes2 in expert_nll.hip
+  int64_t bytes = a16i(8 * 10) + a16i(4 * k * sa_of(k)) +
+                  a16i(4 * tsz_of(k)) + a16i(4 * k * (d + 1)) +
+                  4 * a16i(4 * k) + a16i(4 * d) + 16;
   return bytes <= 160 * 1024;
 }
 
@@ -212,8 +220,9 @@ std::vector<torch::Tensor> fused_laplace_newton(torch::Tensor X,
 
 bool fused_laplace_newton_supported(int64_t k, int64_t d) {
   if (k > 128 || d > k || k < 1) return false;
-  const int64_t tsz = std::max<int64_t>(k * 33, 448);
-  int64_t bytes = 80 + 4 * (2 * k * (k + 1) + tsz + 8 * k + d) + 16;
+  // EXACT mirror of lap_lds_bytes in laplace.hip
+  int64_t bytes = a16i(8 * 10) + 2 * a16i(4 * k * sa_of(k)) +
+                  a16i(4 * tsz_of(k)) + 8 * a16i(4 * k) + a16i(4 * d) + 16;
   return bytes <= 160 * 1024;
 }
 

@@ -47,8 +47,9 @@
 #include "linalg_lds.h"
 
 struct NllLds {
-  float* A;     // k * (k+1):  lower K -> L/V -> K^-1 -> W0; upper: Kb cache
-  float* T;     // temp: max(k*33, 32*(k+1))
+  float* A;     // k * SA (SA = k+1 up-aligned to 4): lower K -> L/V ->
+                // K^-1 -> W0; upper: Kb cache
+  float* T;     // temp: max(k*36, 32*SA, 448)
   float* X;     // k * (d+1) raw features
   float* yb;    // k
   float* alpha; // k
@@ -60,33 +61,48 @@ struct NllLds {
   int* bad;     // 1
 };
 
+// Every region is padded to 16 B so the vectorized dots can assume
+// aligned bases.  Keep in EXACT sync with the mirror in bindings.cpp.
+static __host__ __device__ inline size_t nll_sa(int k) {
+  return (size_t)((k + 4) & ~3);                       // k+1 up to mult 4
+}
+
+static __host__ __device__ inline size_t nll_tsz(int k) {
+  size_t t = (size_t)k * 36;
+  if (t < 32 * nll_sa(k)) t = 32 * nll_sa(k);
+  if (t < 448) t = 448;
+  return t;
+}
+
+static __host__ __device__ inline size_t a16(size_t n) {
+  return (n + 15) & ~(size_t)15;
+}
+
 static __host__ __device__ inline size_t nll_lds_bytes2(int k, int d) {
-  size_t tsz = (size_t)(k * 33 > 32 * (k + 1) ? k * 33 : 32 * (k + 1));
   size_t off = 0;
-  off += sizeof(double) * 8 + sizeof(double) * 2;      // red + misc
-  off += sizeof(float) * (size_t)k * (k + 1);          // A
-  off += sizeof(float) * tsz;                          // T
-  off += sizeof(float) * (size_t)k * (d + 1);          // X
-  off += sizeof(float) * 4 * k;                        // yb alpha tvec rrow
-  off += sizeof(float) * d;                            // s2
-  off += sizeof(int) * 4;                              // bad (+pad)
+  off += a16(sizeof(double) * 10);                     // red + misc
+  off += a16(sizeof(float) * (size_t)k * nll_sa(k));   // A
+  off += a16(sizeof(float) * nll_tsz(k));              // T
+  off += a16(sizeof(float) * (size_t)k * (d + 1));     // X
+  off += 4 * a16(sizeof(float) * k);                   // yb alpha tvec rrow
+  off += a16(sizeof(float) * d);                       // s2
+  off += 16;                                           // bad (+pad)
   return off;
 }
 
 __device__ inline NllLds carve(char* base, int k, int d) {
   NllLds L;
-  size_t tsz = (size_t)(k * 33 > 32 * (k + 1) ? k * 33 : 32 * (k + 1));
   char* p = base;
-  L.red = (double*)p;   p += sizeof(double) * 8;
-  L.misc = (double*)p;  p += sizeof(double) * 2;
-  L.A = (float*)p;      p += sizeof(float) * (size_t)k * (k + 1);
-  L.T = (float*)p;      p += sizeof(float) * tsz;
-  L.X = (float*)p;      p += sizeof(float) * (size_t)k * (d + 1);
-  L.yb = (float*)p;     p += sizeof(float) * k;
-  L.alpha = (float*)p;  p += sizeof(float) * k;
-  L.tvec = (float*)p;   p += sizeof(float) * k;
-  L.rrow = (float*)p;   p += sizeof(float) * k;
-  L.s2 = (float*)p;     p += sizeof(float) * d;
+  L.red = (double*)p;   L.misc = L.red + 8;
+  p += a16(sizeof(double) * 10);
+  L.A = (float*)p;      p += a16(sizeof(float) * (size_t)k * nll_sa(k));
+  L.T = (float*)p;      p += a16(sizeof(float) * nll_tsz(k));
+  L.X = (float*)p;      p += a16(sizeof(float) * (size_t)k * (d + 1));
+  L.yb = (float*)p;     p += a16(sizeof(float) * k);
+  L.alpha = (float*)p;  p += a16(sizeof(float) * k);
+  L.tvec = (float*)p;   p += a16(sizeof(float) * k);
+  L.rrow = (float*)p;   p += a16(sizeof(float) * k);
+  L.s2 = (float*)p;     p += a16(sizeof(float) * d);
   L.bad = (int*)p;
   return L;
 }
@@ -109,7 +125,7 @@ fused_expert_nll_kernel(const float* __restrict__ Xg,
     out_clk[(size_t)blockIdx.x * 20 + (n)] = wall_clock64(); } while (0)
   PH(0);
   NllLds S = carve(lds_raw, k, d);
-  const int SA = k + 1;
+  const int SA = (int)nll_sa(k);
   const int dp = d + 1;
   const int e = blockIdx.x;
   const int tid = threadIdx.x;
@@ -177,10 +193,10 @@ PH(3);
   PH(4);
     // ---- E: alpha = V^T (V y), y.alpha ------------------------------
   for (int i = tid; i < k; i += WG)
-    S.tvec[i] = dot4(S.A + (size_t)i * SA, 1, S.yb, 1, 0, i + 1);
+    S.tvec[i] = dotv(S.A + (size_t)i * SA, S.yb, 0, i + 1);
   __syncthreads();
   for (int a = tid; a < k; a += WG)
-    S.alpha[a] = dot4(S.A + a, SA, S.tvec, 1, a, k);
+    S.alpha[a] = dotm(S.tvec, S.A + a, SA, a, k);
   __syncthreads();
   double part = 0.0;
   for (int i = tid; i < k; i += WG)
@@ -198,13 +214,13 @@ PH(5);
       const int r = f / ncol, j = f - r * ncol;
       const int i = ib + r;
       if (j > i) continue;
-      S.T[r * (k + 1) + j] = dot4(S.A + i, SA, S.A + j, SA, i, k);
+      S.T[r * SA + j] = dot4(S.A + i, SA, S.A + j, SA, i, k);
     }
     __syncthreads();
     for (int f = tid; f < bs * ncol; f += WG) {
       const int r = f / ncol, j = f - r * ncol;
       if (j > ib + r) continue;
-      S.A[(size_t)(ib + r) * SA + j] = S.T[r * (k + 1) + j];
+      S.A[(size_t)(ib + r) * SA + j] = S.T[r * SA + j];
     }
     __syncthreads();
   }
@@ -253,7 +269,7 @@ PH(8);
     const int dl = min(32, d - d0);
     for (int f = tid; f < k * dl; f += WG) {
       const int a = f / dl, j = f - a * dl;
-      S.T[a * 33 + j] = dot4(S.A + (size_t)a * SA, 1,
+      S.T[a * 36 + j] = dotm(S.A + (size_t)a * SA,
                              S.X + d0 + j, dp, 0, k);
     }
     __syncthreads();
@@ -263,7 +279,7 @@ PH(8);
       for (int a = 0; a < k; ++a) {
         const float x = S.X[a * dp + d0 + j];
         acc += 2.0 * (double)x *
-               ((double)x * (double)S.rrow[a] - (double)S.T[a * 33 + j]);
+               ((double)x * (double)S.rrow[a] - (double)S.T[a * 36 + j]);
       }
       out_contr[(size_t)e * d + d0 + j] = acc;
     }

@@ -31,9 +31,9 @@
 #include "linalg_lds.h"
 
 struct LapLds {
-  float* KB;    // k * (k+1): full K = amp*Kb + noise*I (both triangles)
-  float* A;     // k * (k+1): X staging, then B -> L -> V per iteration
-  float* T;     // max(k*33, 448) scratch for the chol machinery
+  float* KB;    // k * SA: full K = amp*Kb + noise*I (both triangles)
+  float* A;     // k * SA: X staging, then B -> L -> V per iteration
+  float* T;     // max(k*36, 32*SA, 448) scratch for the chol machinery
   float* yb;    // k
   float* fb;    // k   latent
   float* pi;    // k
@@ -48,34 +48,50 @@ struct LapLds {
   int* bad;     // 1
 };
 
+// Regions padded to 16 B (vectorized dots assume aligned bases); keep in
+// EXACT sync with the mirror in bindings.cpp.
+static __host__ __device__ inline size_t lap_sa(int k) {
+  return (size_t)((k + 4) & ~3);                       // k+1 up to mult 4
+}
+
+static __host__ __device__ inline size_t lap_tsz(int k) {
+  size_t t = (size_t)k * 36;
+  if (t < 32 * lap_sa(k)) t = 32 * lap_sa(k);
+  if (t < 448) t = 448;
+  return t;
+}
+
+static __host__ __device__ inline size_t lap_a16(size_t n) {
+  return (n + 15) & ~(size_t)15;
+}
+
 static __host__ __device__ inline size_t lap_lds_bytes(int k, int d) {
-  size_t tsz = (size_t)(k * 33 > 448 ? k * 33 : 448);
-  size_t off = sizeof(double) * 10;
-  off += sizeof(float) * (size_t)k * (k + 1) * 2;   // KB + A
-  off += sizeof(float) * tsz;
-  off += sizeof(float) * 8 * k + sizeof(float) * d;
+  size_t off = lap_a16(sizeof(double) * 10);
+  off += 2 * lap_a16(sizeof(float) * (size_t)k * lap_sa(k));  // KB + A
+  off += lap_a16(sizeof(float) * lap_tsz(k));
+  off += 8 * lap_a16(sizeof(float) * k);
+  off += lap_a16(sizeof(float) * d);
   off += 16;
   return off;
 }
 
 __device__ inline LapLds lap_carve(char* base, int k, int d) {
   LapLds L;
-  size_t tsz = (size_t)(k * 33 > 448 ? k * 33 : 448);
   char* p = base;
-  L.red = (double*)p;  p += sizeof(double) * 8;
-  L.misc = (double*)p; p += sizeof(double) * 2;
-  L.KB = (float*)p;    p += sizeof(float) * (size_t)k * (k + 1);
-  L.A = (float*)p;     p += sizeof(float) * (size_t)k * (k + 1);
-  L.T = (float*)p;     p += sizeof(float) * tsz;
-  L.yb = (float*)p;    p += sizeof(float) * k;
-  L.fb = (float*)p;    p += sizeof(float) * k;
-  L.pi = (float*)p;    p += sizeof(float) * k;
-  L.sqw = (float*)p;   p += sizeof(float) * k;
-  L.bv = (float*)p;    p += sizeof(float) * k;
-  L.av = (float*)p;    p += sizeof(float) * k;
-  L.t1 = (float*)p;    p += sizeof(float) * k;
-  L.t2 = (float*)p;    p += sizeof(float) * k;
-  L.s2 = (float*)p;    p += sizeof(float) * d;
+  L.red = (double*)p;  L.misc = L.red + 8;
+  p += lap_a16(sizeof(double) * 10);
+  L.KB = (float*)p;    p += lap_a16(sizeof(float) * (size_t)k * lap_sa(k));
+  L.A = (float*)p;     p += lap_a16(sizeof(float) * (size_t)k * lap_sa(k));
+  L.T = (float*)p;     p += lap_a16(sizeof(float) * lap_tsz(k));
+  L.yb = (float*)p;    p += lap_a16(sizeof(float) * k);
+  L.fb = (float*)p;    p += lap_a16(sizeof(float) * k);
+  L.pi = (float*)p;    p += lap_a16(sizeof(float) * k);
+  L.sqw = (float*)p;   p += lap_a16(sizeof(float) * k);
+  L.bv = (float*)p;    p += lap_a16(sizeof(float) * k);
+  L.av = (float*)p;    p += lap_a16(sizeof(float) * k);
+  L.t1 = (float*)p;    p += lap_a16(sizeof(float) * k);
+  L.t2 = (float*)p;    p += lap_a16(sizeof(float) * k);
+  L.s2 = (float*)p;    p += lap_a16(sizeof(float) * d);
   L.bad = (int*)p;
   return L;
 }
@@ -99,7 +115,7 @@ fused_laplace_newton_kernel(const float* __restrict__ Xg,   // [E, k, d]
                             int* __restrict__ out_iters,       // [E]
                             int* __restrict__ out_bad) {       // [E]
   extern __shared__ char lds_raw[];
-  const int SA = k + 1;
+  const int SA = (int)lap_sa(k);
   LapLds S = lap_carve(lds_raw, k, d);
   const int e = blockIdx.x;
   const int tid = threadIdx.x;
@@ -165,7 +181,7 @@ fused_laplace_newton_kernel(const float* __restrict__ Xg,   // [E, k, d]
     __syncthreads();
     // t1 = K b
     for (int i = tid; i < k; i += WG)
-      S.t1[i] = dot4(S.KB + (size_t)i * SA, 1, S.bv, 1, 0, k);
+      S.t1[i] = dotv(S.KB + (size_t)i * SA, S.bv, 0, k);
     __syncthreads();
     // u = sqw * (K b) into t1; build B (lower) into A
     for (int i = tid; i < k; i += WG) S.t1[i] *= S.sqw[i];
@@ -183,10 +199,10 @@ fused_laplace_newton_kernel(const float* __restrict__ Xg,   // [E, k, d]
     if (*S.bad) break;
     // z = V u (lower); t2 = V^T z
     for (int i = tid; i < k; i += WG)
-      S.t2[i] = dot4(S.A + (size_t)i * SA, 1, S.t1, 1, 0, i + 1);
+      S.t2[i] = dotv(S.A + (size_t)i * SA, S.t1, 0, i + 1);
     __syncthreads();
     for (int a = tid; a < k; a += WG)
-      S.t1[a] = dot4(S.A + a, SA, S.t2, 1, a, k);
+      S.t1[a] = dotm(S.t2, S.A + a, SA, a, k);
     __syncthreads();
     // a = b - sqw * t1
     for (int i = tid; i < k; i += WG)
@@ -194,7 +210,7 @@ fused_laplace_newton_kernel(const float* __restrict__ Xg,   // [E, k, d]
     __syncthreads();
     // t1 = K a;  f_cand (t2) = (1-s) f + s K a
     for (int i = tid; i < k; i += WG)
-      S.t1[i] = dot4(S.KB + (size_t)i * SA, 1, S.av, 1, 0, k);
+      S.t1[i] = dotv(S.KB + (size_t)i * SA, S.av, 0, k);
     __syncthreads();
     const float sf = (float)step;
     double part = 0.0;

@@ -41,6 +41,47 @@ __device__ inline void tri_decode(int f, int& a, int& b) {
   b = f - a * (a + 1) / 2;
 }
 
+// Vectorized contiguous-contiguous dot: both pointers 16-B aligned at
+// index 0 (callers pad their LDS row strides to multiples of 4 floats);
+// scalar head to a 4-aligned c, float4 (ds_read_b128) body, scalar tail.
+__device__ inline float dotv(const float* p, const float* q, int c0,
+                             int c1) {
+  float s = 0.f;
+  int c = c0;
+  for (; c < c1 && (c & 3); ++c) s += p[c] * q[c];
+  float4 a4 = {0.f, 0.f, 0.f, 0.f};
+  for (; c + 3 < c1; c += 4) {
+    const float4 a = *(const float4*)(p + c);
+    const float4 b = *(const float4*)(q + c);
+    a4.x += a.x * b.x;
+    a4.y += a.y * b.y;
+    a4.z += a.z * b.z;
+    a4.w += a.w * b.w;
+  }
+  s += (a4.x + a4.y) + (a4.z + a4.w);
+  for (; c < c1; ++c) s += p[c] * q[c];
+  return s;
+}
+
+// Vectorized-p x strided-q dot: p 16-B aligned at index 0, q any stride.
+__device__ inline float dotm(const float* p, const float* q, int sq,
+                             int c0, int c1) {
+  float s = 0.f;
+  int c = c0;
+  for (; c < c1 && (c & 3); ++c) s += p[c] * q[c * sq];
+  float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
+  for (; c + 3 < c1; c += 4) {
+    const float4 a = *(const float4*)(p + c);
+    s0 += a.x * q[c * sq];
+    s1 += a.y * q[(c + 1) * sq];
+    s2 += a.z * q[(c + 2) * sq];
+    s3 += a.w * q[(c + 3) * sq];
+  }
+  s += (s0 + s1) + (s2 + s3);
+  for (; c < c1; ++c) s += p[c] * q[c * sq];
+  return s;
+}
+
 // 4-accumulator strided dot over LDS: sum_{c=c0}^{c1-1} p[c*sp] * q[c*sq]
 __device__ inline float dot4(const float* p, int sp, const float* q, int sq,
                              int c0, int c1) {
@@ -60,7 +101,8 @@ __device__ inline float dot4(const float* p, int sp, const float* q, int sq,
 // In place on the lower triangle of Abuf (k x k, row stride SA):
 // K -> L -> V = L^{-1} (lower).  Strict upper of Abuf is never touched.
 // log|K| accumulates into misc[0]; *bad set to 1 (indefinite) or 2
-// (non-finite pivot) on breakdown.  Tbuf: >= max(k*33, 448) floats scratch.
+// (non-finite pivot) on breakdown.  Tbuf: >= max(k*36, 448) floats
+// scratch, 16-B aligned; SA must be a multiple of 4 (vectorized dots).
 // All WG threads must call (contains __syncthreads).
 __device__ inline void chol_invert_lower(float* Abuf, float* Tbuf,
                                          const int k, const int SA,
@@ -95,8 +137,8 @@ __device__ inline void chol_invert_lower(float* Abuf, float* Tbuf,
         if (c > r) continue;                 // keep the Kb upper cache
         const int i = jb + r, cc = jb + c;
         Abuf[(size_t)i * SA + cc] -=
-            dot4(Abuf + (size_t)i * SA + pj, 1,
-                 Abuf + (size_t)cc * SA + pj, 1, 0, NB);
+            dotv(Abuf + (size_t)i * SA + pj,
+                 Abuf + (size_t)cc * SA + pj, 0, NB);
       }
       __syncthreads();
     }
@@ -220,8 +262,8 @@ __device__ inline void chol_invert_lower(float* Abuf, float* Tbuf,
           const int r = f / bs, c = f - r * bs;
           const int i = t0r + r, cc = jb + c;
           Abuf[(size_t)i * SA + cc] -=
-              dot4(Abuf + (size_t)i * SA + pj, 1,
-                   Abuf + (size_t)cc * SA + pj, 1, 0, NB);
+              dotv(Abuf + (size_t)i * SA + pj,
+                   Abuf + (size_t)cc * SA + pj, 0, NB);
         }
         for (int f = (tid - 64) + q * (WG - 64); f < ntri_rest;
              f += (WG - 64) * nq) {
@@ -229,8 +271,8 @@ __device__ inline void chol_invert_lower(float* Abuf, float* Tbuf,
           tri_decode(f, a, b);
           const int i = t0r + a, c = t0r + b;
           Abuf[(size_t)i * SA + c] -=
-              dot4(Abuf + (size_t)i * SA + pj, 1,
-                   Abuf + (size_t)c * SA + pj, 1, 0, NB);
+              dotv(Abuf + (size_t)i * SA + pj,
+                   Abuf + (size_t)c * SA + pj, 0, NB);
         }
       }
       __syncthreads();
@@ -262,7 +304,7 @@ __device__ inline void chol_invert_lower(float* Abuf, float* Tbuf,
           tri_decode(f, a, b);
           const int i = p0 + a, c = p0 + b;
           Abuf[(size_t)(jb + i) * SA + jb + c] -=
-              dot4(D + (size_t)i * SA + qb, 1, D + (size_t)c * SA + qb, 1,
+              dotv(D + (size_t)i * SA + qb, D + (size_t)c * SA + qb,
                    0, sbs);
         }
         __syncthreads();
@@ -288,7 +330,7 @@ __device__ inline void chol_invert_lower(float* Abuf, float* Tbuf,
           float u = 0.f;
           if (gi < bs && jb8 + j < bs) {
             // u = sum_{c=jb8+8}^{gi} V[gi][c] * L[c][jb8+j]
-            u = dot4(D + (size_t)gi * SA, 1, D + jb8 + j, SA,
+            u = dotm(D + (size_t)gi * SA, D + jb8 + j, SA,
                      jb8 + 8, min(gi + 1, bs));
           }
           TS[f] = u;
@@ -330,17 +372,17 @@ __device__ inline void chol_invert_lower(float* Abuf, float* Tbuf,
     const int t0 = jb + bs;        // first trailing row
     const int nr = k - t0;         // panel rows
     if (nr > 0) {
-      // C2: copy panel below the diag block into T (row r-t0, stride 33)
+      // C2: copy panel below the diag block into T (row r-t0, stride 36)
       for (int f = tid; f < nr * bs; f += WG) {
         int r = f / bs, c = f - r * bs;
-        Tbuf[r * 33 + c] = Abuf[(size_t)(t0 + r) * SA + jb + c];
+        Tbuf[r * 36 + c] = Abuf[(size_t)(t0 + r) * SA + jb + c];
       }
       __syncthreads();
       // C2b: panel <- T * V_JJ^T : A[r][jb+c] = sum_{t<=c} T[r][t] V[c][t]
       for (int f = tid; f < nr * bs; f += WG) {
         int r = f / bs, c = f - r * bs;
         Abuf[(size_t)(t0 + r) * SA + jb + c] =
-            dot4(Tbuf + r * 33, 1, D + c * SA, 1, 0, c + 1);
+            dotv(Tbuf + r * 36, D + c * SA, 0, c + 1);
       }
       __syncthreads();
     }
@@ -361,15 +403,15 @@ __device__ inline void chol_invert_lower(float* Abuf, float* Tbuf,
     for (int f = tid; f < nr * bs; f += WG) {
       int r = f / bs, t = f - r * bs;
       const int row = t0 + r;
-      Tbuf[r * 33 + t] = dot4(Abuf + (size_t)row * SA, 1,
-                             Abuf + jb + t, SA, t0, row + 1);
+      Tbuf[r * 36 + t] = dotm(Abuf + (size_t)row * SA,
+                              Abuf + jb + t, SA, t0, row + 1);
     }
     __syncthreads();
     // V[row][jb+j] = - sum_{t>=j} U[r][t] * V_JJ[t][j]
     for (int f = tid; f < nr * bs; f += WG) {
       int r = f / bs, j = f - r * bs;
       Abuf[(size_t)(t0 + r) * SA + jb + j] =
-          -dot4(Tbuf + r * 33, 1, Abuf + (size_t)jb * SA + jb + j, SA, j, bs);
+          -dotm(Tbuf + r * 36, Abuf + (size_t)jb * SA + jb + j, SA, j, bs);
     }
     __syncthreads();
   }
