@@ -196,3 +196,51 @@ def test_scaling_distributed_matches_single():
     ref = np.concatenate([s.mean, s.scale])
     np.testing.assert_allclose(results[0], ref, rtol=1e-12)
     np.testing.assert_allclose(results[1], ref, rtol=1e-12)
+
+
+# ---- distributed GPC --------------------------------------------------------
+
+def _w_gpc_fit(rank, world):
+    from spark_gp_amd.models.classification import GaussianProcessClassifier
+    from spark_gp_amd.kernels import ARDRBFKernel
+    rng = np.random.default_rng(7)
+    X = rng.uniform(size=(800, 3))
+    y = (np.sin(4 * X.sum(-1)) > 0).astype(np.float64)
+    Xs, ys = _shard(X, y, rank, world)
+    model = (GaussianProcessClassifier()
+             .setKernel(lambda: 1 * ARDRBFKernel(3))
+             .setDatasetSizeForExpert(50).setActiveSetSize(100)
+             .setSigma2(1e-3).setMaxIter(20).setSeed(0).setDevice("cpu")
+             .fit(Xs, ys))
+    proba = model.predict_proba(X[:100])
+    acc = float((model.predict(X) == y).mean())
+    return proba[:, 1], acc
+
+
+def test_distributed_gpc_agrees_across_ranks_and_single():
+    """Distributed binary GPC (Laplace + allreduced evidence, C1) must give
+    identical models on both ranks and match the single-process fit."""
+    res = _spawn("_w_gpc_fit")
+    p0, acc0 = res[0]
+    p1, acc1 = res[1]
+    np.testing.assert_allclose(p0, p1, atol=1e-10)   # replicated optimizer
+    assert acc0 == acc1
+
+    from spark_gp_amd.models.classification import GaussianProcessClassifier
+    from spark_gp_amd.kernels import ARDRBFKernel
+    rng = np.random.default_rng(7)
+    X = rng.uniform(size=(800, 3))
+    y = (np.sin(4 * X.sum(-1)) > 0).astype(np.float64)
+    single = (GaussianProcessClassifier()
+              .setKernel(lambda: 1 * ARDRBFKernel(3))
+              .setDatasetSizeForExpert(50).setActiveSetSize(100)
+              .setSigma2(1e-3).setMaxIter(20).setSeed(0).setDevice("cpu")
+              .fit(X, y))
+    # expert grouping differs between 1-shard and 2-shard runs, so the
+    # fitted hyperparameters (and probabilities near the decision
+    # boundary) differ slightly; the decision FUNCTION must agree on the
+    # vast majority of points and both fits must classify well
+    ps = single.predict_proba(X[:100])[:, 1]
+    agree = float(((p0 > 0.5) == (ps > 0.5)).mean())
+    assert agree >= 0.95, agree
+    assert acc0 > 0.9
