@@ -23,7 +23,6 @@ import json
 import os
 import time
 
-import numpy as np
 import torch
 
 

@@ -13,7 +13,6 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import argparse
 
-import numpy as np
 
 from spark_gp_amd import (GaussianProcessClassifier, RBFKernel, StandardScaler,
                           accuracy, train_validation_split)

@@ -8,7 +8,6 @@ These run on CPU here; the same code path runs over RCCL/xGMI on MI355X
 import os
 
 import numpy as np
-import pytest
 import torch
 import torch.multiprocessing as mp
 

@@ -1,7 +1,6 @@
 """GPU numerics tests: every HIP kernel against the plain-PyTorch fp32/fp64
 reference of the same op (required test shape — see repo instructions)."""
 
-import math
 
 import numpy as np
 import pytest
@@ -241,7 +240,6 @@ def test_gpu_fit_mixed_precision_ppa(dev, ext):
 
 
 def test_gpu_classifier_end_to_end(dev, ext):
-    import os
     from spark_gp_amd import GaussianProcessClassifier, RBFKernel, accuracy
     rng = np.random.default_rng(0)
     n = 4000

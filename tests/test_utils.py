@@ -2,7 +2,6 @@
 memoized objective."""
 
 import math
-import os
 
 import numpy as np
 import pytest
