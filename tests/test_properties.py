@@ -83,8 +83,9 @@ def test_compiled_canonical_matches_tree(amp, beta, noise, seed):
     s = torch.as_tensor(theta[cs.base_idx])
     Xs = X * s
     from spark_gp_amd.kernels import sqdist
-    canon = (C * torch.exp(-sqdist(Xs, Xs)) + nu * torch.eye(6)).numpy()
-    np.testing.assert_allclose(direct, canon, rtol=1e-9, atol=1e-12)
+    canon = (C * torch.exp(-sqdist(Xs, Xs))
+             + nu * torch.eye(6, dtype=X.dtype)).numpy()
+    np.testing.assert_allclose(direct, canon, rtol=1e-12, atol=1e-14)
 
 
 def test_refit_is_deterministic():
