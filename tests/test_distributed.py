@@ -243,3 +243,38 @@ def test_distributed_gpc_agrees_across_ranks_and_single():
     agree = float(((p0 > 0.5) == (ps > 0.5)).mean())
     assert agree >= 0.95, agree
     assert acc0 > 0.9
+
+
+# ---- distributed greedy active set -----------------------------------------
+
+def _w_greedy(rank, world):
+    from spark_gp_amd import GreedilyOptimizingActiveSetProvider
+    from spark_gp_amd.kernels import ARDRBFKernel, EyeKernel, Scalar
+    from spark_gp_amd.parallel.dist import get_comm
+    X, y = performance_benchmark_data(300, 3, seed=11, dtype=np.float64)
+    y = np.sin(X.sum(-1) * 3.0)
+    Xl, yl = _shard(X, y, rank, world)
+    kernel = 1 * ARDRBFKernel(3) + Scalar(1e-2).const * EyeKernel()
+    prov = GreedilyOptimizingActiveSetProvider()
+    active = prov(12, torch.tensor(Xl), torch.tensor(yl), kernel,
+                  kernel.get_hyperparameters(), 5, get_comm())
+    return active.numpy()
+
+
+def test_greedy_provider_distributed_matches_single():
+    """The Seeger forward selection's per-round collectives (C2-style stat
+    allreduce + C6/C7 distributed argmax) must reproduce the single-process
+    selection exactly: same seed, same global data => same active set on
+    every rank."""
+    results = _spawn("_w_greedy")
+    from spark_gp_amd import GreedilyOptimizingActiveSetProvider
+    from spark_gp_amd.kernels import ARDRBFKernel, EyeKernel, Scalar
+    from spark_gp_amd.parallel.dist import Comm
+    X, y = performance_benchmark_data(300, 3, seed=11, dtype=np.float64)
+    y = np.sin(X.sum(-1) * 3.0)
+    kernel = 1 * ARDRBFKernel(3) + Scalar(1e-2).const * EyeKernel()
+    ref = GreedilyOptimizingActiveSetProvider()(
+        12, torch.tensor(X), torch.tensor(y), kernel,
+        kernel.get_hyperparameters(), 5, Comm()).numpy()
+    np.testing.assert_allclose(results[0], results[1], atol=0)
+    np.testing.assert_allclose(results[0], ref, atol=1e-12)
