@@ -54,6 +54,30 @@ def test_fused_expert_nll_ard_vs_oracle(dev, ext):
                                atol=2e-3 * np.abs(grad_o).max())
 
 
+@pytest.mark.parametrize("k,d", [(13, 5), (24, 7), (7, 3), (31, 4)])
+def test_fused_expert_nll_small_odd_shapes(dev, ext, k, d):
+    """Partial 32-blocks, partial 8x8 sub-blocks and odd (non-multiple-of-4)
+    k exercise every tail path of the blocked factorization and the aligned
+    LDS strides."""
+    from spark_gp_amd.kernels import (ARDRBFKernel, EyeKernel, Scalar,
+                                      compile_kernel)
+    from spark_gp_amd.ops import hip_backend, torch_backend
+    E = 6
+    X, y = _ard_setup(E, k, d, seed=5, dev=dev)
+    cs = compile_kernel(1 * ARDRBFKernel(d) + Scalar(1e-3).const * EyeKernel())
+    rng = np.random.default_rng(2)
+    theta = np.concatenate([[0.9], rng.uniform(0.5, 2.0, d)])
+    scale = torch.as_tensor(theta[1:], dtype=torch.float32, device=dev)
+    *_, bad = ext.fused_expert_nll(X, y, scale, float(theta[0]), 1e-3)
+    assert int(bad.sum()) == 0, "experts fell back: kernel not exercised"
+    nll_h, grad_h = hip_backend.nll_grad_compiled(cs, theta, X, y)
+    nll_o, grad_o = torch_backend.nll_grad_compiled(
+        cs, theta, X.double().cpu(), y.double().cpu())
+    assert nll_h == pytest.approx(nll_o, rel=2e-4)
+    np.testing.assert_allclose(grad_h, grad_o, rtol=3e-3,
+                               atol=2e-3 * np.abs(grad_o).max())
+
+
 def test_fused_expert_nll_rbf_vs_oracle(dev, ext):
     from spark_gp_amd.kernels import (EyeKernel, RBFKernel, Scalar,
                                       WhiteNoiseKernel, compile_kernel)
