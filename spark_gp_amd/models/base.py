@@ -62,9 +62,32 @@ def group_experts(X: torch.Tensor, y: torch.Tensor, k_target: int
 
 
 class GaussianProcessParams:
-    """Fluent parameter mixin with the reference's defaults."""
+    """Fluent parameter mixin with the reference's defaults.
 
-    def __init__(self):
+    Also sklearn-compatible: ``get_params``/``set_params``/keyword
+    construction follow the scikit-learn estimator contract, so
+    ``sklearn.base.clone`` and tools like ``cross_val_score`` work (the
+    estimator additionally keeps its last fitted model in ``model_`` and
+    delegates ``predict`` to it for those tools; the primary API remains
+    Spark-style ``fit(X, y) -> Model``)."""
+
+    # sklearn param name -> backing attribute
+    _PARAMS = {
+        "kernel": "_kernel_factory",
+        "dataset_size_for_expert": "_dataset_size_for_expert",
+        "sigma2": "_sigma2",
+        "active_set_size": "_active_set_size",
+        "active_set_provider": "_active_set_provider",
+        "max_iter": "_max_iter",
+        "tol": "_tol",
+        "seed": "_seed",
+        "device": "_device",
+        "dtype": "_dtype",
+        "ppa_precision": "_ppa_precision",
+        "optimizer_restart": "_optimizer_restart",
+    }
+
+    def __init__(self, **params):
         self._kernel_factory: Callable[[], Kernel] = lambda: RBFKernel()
         self._dataset_size_for_expert = 100
         self._sigma2 = 1e-3
@@ -78,6 +101,29 @@ class GaussianProcessParams:
         self._ppa_precision = "fp64"  # 'fp64' (reference-parity) | 'mixed'
                                       # (hi/lo bf16 MFMA SYRK, fastest)
         self._optimizer_restart = True  # restart-on-bound-collapse guard
+        self.model_ = None            # last fitted model (sklearn interop)
+        if params:
+            self.set_params(**params)
+
+    # sklearn estimator contract --------------------------------------
+    def get_params(self, deep: bool = True):
+        return {name: getattr(self, attr) for name, attr in self._PARAMS.items()}
+
+    def set_params(self, **params):
+        for name, value in params.items():
+            attr = self._PARAMS.get(name)
+            if attr is None:
+                raise ValueError(f"unknown parameter {name!r}; valid: "
+                                 f"{sorted(self._PARAMS)}")
+            setattr(self, attr, value)
+        return self
+
+    def predict(self, X, **kw):
+        """Delegate to the last fitted model (sklearn tools call
+        ``est.fit(...); est.predict(...)`` on the estimator itself)."""
+        if self.model_ is None:
+            raise RuntimeError("estimator is not fitted; call fit(X, y)")
+        return self.model_.predict(X, **kw)
 
     # Reference-parity camelCase setters -------------------------------
     def setKernel(self, factory: Callable[[], Kernel]):

@@ -23,9 +23,12 @@ from .predictor import GaussianProjectedProcessRawPredictor
 
 
 class GaussianProcessClassifier(GaussianProcessCommons):
-    def __init__(self):
-        super().__init__()
+    _PARAMS = dict(GaussianProcessCommons._PARAMS,
+                   max_newton_iter="_max_newton_iter")
+
+    def __init__(self, **params):
         self._max_newton_iter = 200
+        super().__init__(**params)
 
     def setMaxNewtonIter(self, v: int):
         self._max_newton_iter = int(v)
@@ -71,6 +74,7 @@ class GaussianProcessClassifier(GaussianProcessCommons):
         instr.log_success()
         model = GaussianProcessClassificationModel(raw)
         model._instr = instr
+        self.model_ = model           # sklearn-interop handle
         return model
 
 

@@ -54,6 +54,7 @@ class GaussianProcessRegression(GaussianProcessCommons):
         instr.log_success()
         model = GaussianProcessRegressionModel(raw)
         model._instr = instr
+        self.model_ = model           # sklearn-interop handle
         return model
 
 
