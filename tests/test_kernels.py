@@ -161,3 +161,18 @@ def test_sqdist_nonnegative_and_symmetric():
     assert (sq >= 0).all()
     np.testing.assert_allclose(sq.numpy(), sq.T.numpy(), atol=1e-12)
     assert np.allclose(np.diag(sq.numpy()), 0.0, atol=1e-12)
+
+
+def test_rbf_reference_fixture_parity():
+    """The reference's own unit-test fixture (``RBFKernelTest.scala:27-39``):
+    dataset [(1,2), (2,3), (5,7)], sigma = sqrt(0.2), expected K to 1e-4 —
+    numeric parity with the reference's asserted values."""
+    X = torch.tensor([[1.0, 2.0], [2.0, 3.0], [5.0, 7.0]], dtype=torch.float64)
+    k = RBFKernel(math.sqrt(0.2))
+    K = k.training_kernel(X).numpy()
+    expected = np.array([[1.000000e+00, 6.737947e-03, 3.053624e-45],
+                         [6.737947e-03, 1.000000e+00, 7.187782e-28],
+                         [3.053624e-45, 7.187782e-28, 1.000000e+00]])
+    np.testing.assert_allclose(K, expected, atol=1e-4)
+    # and much tighter on the representable entries
+    np.testing.assert_allclose(K[0, 1], expected[0, 1], rtol=1e-6)
