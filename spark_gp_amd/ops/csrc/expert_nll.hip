@@ -11,14 +11,16 @@
 //   B  lower(A) = amp*Kb + noise*I,  strict upper(A) = Kb (cached base
 //      kernel — read back in phase W, never recomputed)
 //   C  blocked right-looking Cholesky, IN PLACE, with look-ahead: per
-//      32-column block, wave 0 factors AND inverts the diagonal block
-//      (8x8 sub-diagonals factored+inverted entirely inside lane 0's
-//      registers — the serial dependency chain never touches LDS — with
-//      lane-parallel 8-wide panels/trailing and block back-substitution
-//      assembling the 32x32 inverse) WHILE waves 1-7 apply the previous
-//      panel's rank-32 trailing update to the remaining columns; the
-//      panel solve is a dense GEMM against the inverted diagonal.
-//      fp32; fp64 logdet.
+//      8-column sub-block q of each 32-column block, wave 0 factors AND
+//      inverts the 8x8 sub-diagonal (row-per-lane on 8 lanes, cross-lane
+//      traffic via __shfl at STATIC register indices; branchless
+//      frexp-accumulated log-det) WHILE waves 1-7 apply chunk q of the
+//      previous panel's rank-32 trailing update; then ALL 512 threads do
+//      the intra-block panel/trailing and, after the 4 sub-blocks, the
+//      block back-substitution assembling the 32x32 inverse.  The panel
+//      solve is a dense GEMM against the inverted diagonal.
+//      fp32; fp64 logdet.  (Design history and negative results:
+//      profiles/PROFILES.md and TODO.md item 1.)
 //   D  blocked in-place triangular inverse of the off-diagonal blocks
 //      (right-to-left column blocks, rows fully parallel:
 //       V_IJ = -(sum_{K>J} V_IK L_KJ) L_JJ^-1).
@@ -30,9 +32,11 @@
 //   H  gradient contraction per input dim (the K5 fusion):
 //      contr_j = 2 sum_a x_aj^2 r_a - 2 sum_a x_aj (W0 X)_aj
 //
-// All serial inner products are multi-accumulator-unrolled so they run at
-// LDS throughput instead of FMA-latency.  Single k x (k+1) working buffer +
-// one k x 33 temp => ~69 KB LDS at k=100, two experts resident per CU.
+// Inner products with a contiguous operand are float4 (ds_read_b128)
+// vectorized over 16-B-aligned LDS rows (strides padded to multiples of 4
+// floats); the rest are multi-accumulator-unrolled scalars.  Single
+// k x SA working buffer + one k x 36 temp => ~69 KB LDS at k=100, two
+// experts resident per CU.
 //
 // Numerics: fp32 storage/factorization, fp64 scalar accumulation.  Experts
 // whose fp32 Cholesky breaks down are flagged in out_bad and recomputed on
