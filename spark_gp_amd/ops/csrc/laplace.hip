@@ -18,9 +18,11 @@
 // Outputs: updated latent f (in place), psi, sum log diag L, per-expert
 // Newton iteration count, bad flag (fp32 breakdown -> host falls back to
 // the batched torch path for that expert).  The evidence/gradient pass
-// (Algorithm 5.1) stays on the torch path, started from the converged f —
-// its own Newton loop then terminates in 2-3 cheap iterations, preserving
-// the reference's exact evidence semantics.
+// (Algorithm 5.1) runs on the torch path AT the converged f in contraction
+// form (torch_backend.laplace_evidence_compiled — one batched fp64
+// Cholesky, no [E, p, k, k] tensor, reference-exact semantics); if any
+// expert went bad, the whole batch falls back to the torch Newton loop
+// warm-started from f.
 //
 // Constraints: k <= 128, d <= k (X stages through the A buffer).
 
