@@ -278,3 +278,32 @@ def test_greedy_provider_distributed_matches_single():
         kernel.get_hyperparameters(), 5, Comm()).numpy()
     np.testing.assert_allclose(results[0], results[1], atol=0)
     np.testing.assert_allclose(results[0], ref, atol=1e-12)
+
+
+# ---- empty-shard robustness -------------------------------------------------
+
+def _w_empty_shard(rank, world):
+    from spark_gp_amd import GaussianProcessRegression
+    from spark_gp_amd.kernels import ARDRBFKernel, EyeKernel, Scalar
+    X, y = performance_benchmark_data(200, 3, seed=21, dtype=np.float64)
+    y = np.sin(X.sum(-1) * 3.0)
+    # rank 1 gets NO data at all
+    Xl, yl = (X, y) if rank == 0 else (X[:0], y[:0])
+    model = (GaussianProcessRegression()
+             .setKernel(lambda: 1 * ARDRBFKernel(3)
+                        + Scalar(1e-2).const * EyeKernel())
+             .setDatasetSizeForExpert(50).setActiveSetSize(40)
+             .setSigma2(1e-3).setMaxIter(30).setSeed(2).setDevice("cpu")
+             .fit(Xl, yl))
+    return model.predict(X[:30])
+
+
+def test_rank_with_empty_shard_participates():
+    """A rank holding zero rows must still participate in every collective
+    (objective, sampling, PPA) and end with the identical model — uneven
+    data feeds are normal in production ingestion."""
+    res = _spawn("_w_empty_shard")
+    np.testing.assert_allclose(res[0], res[1], atol=1e-12)
+    X, _ = performance_benchmark_data(200, 3, seed=21, dtype=np.float64)
+    yq = np.sin(X[:30].sum(-1) * 3.0)
+    assert float(np.sqrt(np.mean((res[0] - yq) ** 2))) < 0.25
