@@ -135,3 +135,30 @@ def test_predictive_variance_shrinks_at_training_points():
     assert std_far.min() > 3 * std_in.mean()
     # far from data the PPA mean falls back toward the prior mean 0
     assert np.abs(mean_far).max() < 0.1
+
+
+def test_group_experts_partition_properties():
+    """Every row lands in exactly one expert; sizes differ by at most 1;
+    assignment is the reference's round-robin (row i -> expert i % E)
+    re-grouped by size (``GaussianProcessCommons.scala:26-31``)."""
+    import torch
+    from spark_gp_amd.models.base import group_experts
+    for n, kt in [(101, 10), (100, 100), (7, 3), (250, 100), (5, 100)]:
+        X = torch.arange(n, dtype=torch.float64).unsqueeze(-1).repeat(1, 2)
+        y = torch.arange(n, dtype=torch.float64)
+        groups = group_experts(X, y, kt)
+        E = max(1, int(round(n / kt)))
+        all_idx = torch.cat([idx for idx, _, _ in groups])
+        assert sorted(all_idx.tolist()) == list(range(n))
+        sizes = [Xg.shape[1] for _, Xg, _ in groups for _ in range(Xg.shape[0])]
+        assert max(sizes) - min(sizes) <= 1
+        assert sum(sizes) == n
+        # row -> expert is i % E: rows of one expert are congruent mod E
+        for idx, Xg, yg in groups:
+            flat = idx.reshape(Xg.shape[0], Xg.shape[1])
+            for row_ids in flat:
+                assert len(set(int(i) % E for i in row_ids)) == 1
+        # X/y rows carried intact
+        for idx, Xg, yg in groups:
+            assert torch.equal(Xg[..., 0].reshape(-1).to(torch.int64), idx)
+            assert torch.equal(yg.reshape(-1).to(torch.int64), idx)
