@@ -338,7 +338,9 @@ def laplace_evidence_compiled(cs: CompiledKernel, theta: np.ndarray,
     R = sqw.unsqueeze(-1) * torch.cholesky_solve(torch.diag_embed(sqw), L)
     KR = K @ R
     d3 = -(2.0 * pi - 1.0) * pi * pi * torch.exp(-f)
-    diagKRK = ((KR @ K) * eyek).sum(-1)          # (K R K)_ii
+    # (K R K)_ii = sum_j (KR)_ij K_ji = (KR o K).sum(-1) since K is
+    # symmetric — avoids materializing the full [E, k, k] product
+    diagKRK = (KR * K).sum(-1)
     s2 = -0.5 * (K.diagonal(dim1=-2, dim2=-1) - diagKRK) * d3
 
     # shared pieces: s1_i = 1/2 sum(G o dK_i) with G = a a^T - R;
@@ -359,9 +361,11 @@ def laplace_evidence_compiled(cs: CompiledKernel, theta: np.ndarray,
         t1 = (X * X * r.unsqueeze(-1)).sum(-2)
         t2 = (X * (W0 @ X)).sum(-2)
         s1 = -bt * C * (2.0 * t1 - 2.0 * t2)     # [E, d]; dK_j has -2 beta_j
-        u0 = (Kc @ v.unsqueeze(-1)).squeeze(-1)
-        U1 = Kc @ (X * v.unsqueeze(-1))
-        U2 = Kc @ (X * X * v.unsqueeze(-1))
+        # one batched GEMM for all three dK_j-matvec ingredients
+        d = X.shape[-1]
+        U = Kc @ torch.cat([v.unsqueeze(-1), X * v.unsqueeze(-1),
+                            X * X * v.unsqueeze(-1)], dim=-1)
+        u0, U1, U2 = U[..., 0], U[..., 1:1 + d], U[..., 1 + d:]
         Bm = -2.0 * bt * (X * X * u0.unsqueeze(-1) - 2.0 * X * U1 + U2)
         grad[cs.base_idx] = (s1.double().sum(0)
                              + s2_dot_s3(Bm).double().sum(0)).cpu().numpy()
@@ -372,9 +376,10 @@ def laplace_evidence_compiled(cs: CompiledKernel, theta: np.ndarray,
         t2 = (Xs * (W0 @ Xs)).sum(-2)
         # s1 = 1/2 * (2C/sigma) * sum(W0 o sq_scaled)
         s1 = (C / sigma) * (2.0 * t1 - 2.0 * t2).sum(-1)         # [E]
-        u0 = (Kc @ v.unsqueeze(-1)).squeeze(-1)
-        U1 = Kc @ (Xs * v.unsqueeze(-1))
-        U2 = Kc @ (Xs * Xs * v.unsqueeze(-1))
+        d = Xs.shape[-1]
+        U = Kc @ torch.cat([v.unsqueeze(-1), Xs * v.unsqueeze(-1),
+                            Xs * Xs * v.unsqueeze(-1)], dim=-1)
+        u0, U1, U2 = U[..., 0], U[..., 1:1 + d], U[..., 1 + d:]
         Bm = (2.0 / sigma) * ((Xs * Xs).sum(-1) * u0
                               - 2.0 * (Xs * U1).sum(-1) + U2.sum(-1))
         grad[cs.base_idx.start] = float(
