@@ -186,3 +186,29 @@ def test_laplace_evidence_compiled_matches_generic():
         assert nll_c == pytest.approx(nll_ref, rel=1e-6)
         np.testing.assert_allclose(grad_c, grad_ref, rtol=1e-5,
                                    atol=1e-6 * np.abs(grad_ref).max())
+
+
+def test_averaged_proba_shrinks_toward_half():
+    """Gauss-Hermite averaging over the latent posterior must pull
+    probabilities toward 0.5 relative to the plain sigmoid-of-mean
+    (Jensen: E[sigmoid(Z)] is closer to 1/2 than sigmoid(E[Z]) for the
+    logistic link), and agree with it where the posterior is confident."""
+    rng = np.random.default_rng(4)
+    X = np.concatenate([rng.normal(-1.0, 0.6, (200, 2)),
+                        rng.normal(1.0, 0.6, (200, 2))])
+    y = np.concatenate([np.zeros(200), np.ones(200)])
+    model = (GaussianProcessClassifier()
+             .setKernel(lambda: 1 * RBFKernel(1.0, 1e-3, 10))
+             .setDatasetSizeForExpert(50).setActiveSetSize(60)
+             .setSigma2(1e-3).setMaxIter(30).setSeed(0).setDevice("cpu")
+             .fit(X, y))
+    # points far outside the data: high posterior variance
+    Xq = np.concatenate([X[:50], rng.normal(0.0, 6.0, (50, 2))])
+    p_plain = model.predict_proba(Xq)[:, 1]
+    p_avg = model.predict_proba(Xq, averaged=True)[:, 1]
+    # same labels either way
+    assert ((p_plain > 0.5) == (p_avg > 0.5)).all()
+    # averaging never moves AWAY from 1/2 (up to fp noise)
+    assert (np.abs(p_avg - 0.5) <= np.abs(p_plain - 0.5) + 1e-12).all()
+    # and strictly shrinks somewhere on the uncertain points
+    assert (np.abs(p_avg[50:] - 0.5) < np.abs(p_plain[50:] - 0.5) - 1e-6).any()
