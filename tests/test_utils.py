@@ -173,3 +173,26 @@ def test_integrator_averaged_proba_single_matches_batch():
         single = integ.expected_of_function_of_normal(
             f[i], v[i], lambda z: 1.0 / (1.0 + np.exp(-z)))
         assert single == pytest.approx(batch[i], rel=1e-12)
+
+
+def test_kmeans_empty_clusters_keep_previous_centers():
+    """With heavy duplicates, Lloyd iterations empty some clusters; their
+    centers must survive (no NaN division) and all centers stay finite."""
+    import torch
+    from spark_gp_amd import KMeansActiveSetProvider
+    from spark_gp_amd.parallel.dist import Comm
+    X = torch.tensor([[0.0, 0.0]] * 50 + [[10.0, 10.0]] * 50,
+                     dtype=torch.float64)
+    y = torch.zeros(100, dtype=torch.float64)
+    centers = KMeansActiveSetProvider(max_iter=8)(
+        8, X, y, None, None, 3, Comm())
+    assert centers.shape == (8, 2)
+    assert torch.isfinite(centers).all()
+
+
+def test_serve_missing_model_dir_fails_cleanly():
+    import pytest as _pytest
+    _pytest.importorskip("fastapi")
+    from spark_gp_amd.serve import create_app
+    with _pytest.raises(FileNotFoundError):
+        create_app("/nonexistent/model/dir")
