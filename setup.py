@@ -8,7 +8,7 @@ hipcc cross-compiles gfx950 without a GPU present; the resulting
 
 import os
 
-from setuptools import setup
+from setuptools import find_packages, setup
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
@@ -33,7 +33,8 @@ ext = CUDAExtension(
 setup(
     name="spark_gp_amd",
     version="0.1.0",
-    packages=["spark_gp_amd"],
+    packages=find_packages(include=["spark_gp_amd", "spark_gp_amd.*"]),
+    package_data={"spark_gp_amd": ["data/*.csv", "ops/csrc/*"]},
     ext_modules=[ext],
     cmdclass={"build_ext": BuildExtension},
 )
