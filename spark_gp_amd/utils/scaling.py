@@ -42,6 +42,10 @@ class StandardScaler:
     def fit_transform(self, X, comm: Optional[Comm] = None):
         return self.fit(X, comm).transform(X)
 
+    def inverse_transform(self, X):
+        Xn = np.asarray(X, dtype=np.float64)
+        return Xn * self.scale + self.mean
+
 
 def scale(X, y, comm: Optional[Comm] = None) -> Tuple[np.ndarray, np.ndarray]:
     """Convenience mirroring the reference's ``scale(RDD[LabeledPoint])``."""

@@ -196,3 +196,28 @@ def test_serve_missing_model_dir_fails_cleanly():
     from spark_gp_amd.serve import create_app
     with _pytest.raises(FileNotFoundError):
         create_app("/nonexistent/model/dir")
+
+
+def test_scaler_inverse_transform_roundtrip():
+    from spark_gp_amd import StandardScaler
+    rng = np.random.default_rng(8)
+    X = rng.normal(3.0, [1.0, 5.0, 0.0], size=(100, 3))   # one constant dim
+    s = StandardScaler().fit(X)
+    np.testing.assert_allclose(s.inverse_transform(s.transform(X)), X,
+                               rtol=1e-12, atol=1e-12)
+
+
+def test_synthetic_generators_deterministic():
+    from spark_gp_amd.data import (benchmark_regression_data,
+                                   performance_benchmark_data,
+                                   shard_performance_benchmark_data)
+    X1, y1 = performance_benchmark_data(100, 3, seed=7)
+    X2, y2 = performance_benchmark_data(100, 3, seed=7)
+    np.testing.assert_array_equal(X1, X2)
+    np.testing.assert_array_equal(y1, y2)
+    # shards partition the total row count
+    tot = sum(shard_performance_benchmark_data(103, 4, r, 4)[0].shape[0]
+              for r in range(4))
+    assert tot == 103
+    Xb, yb = benchmark_regression_data(50, 4, seed=1)
+    assert Xb.shape == (50, 4) and yb.shape == (50,) and np.isfinite(yb).all()
