@@ -125,6 +125,19 @@ class GaussianProcessParams:
             raise RuntimeError("estimator is not fitted; call fit(X, y)")
         return self.model_.predict(X, **kw)
 
+    def score(self, X, y):
+        """Default sklearn scorer: R^2 for regression, accuracy for
+        classification (decided by the fitted model's kind)."""
+        import numpy as _np
+        pred = self.predict(X)
+        y = _np.asarray(y, dtype=_np.float64).reshape(-1)
+        from .classification import GaussianProcessClassificationModel
+        if isinstance(self.model_, GaussianProcessClassificationModel):
+            return float((pred == y).mean())
+        ss_res = float(((y - pred) ** 2).sum())
+        ss_tot = float(((y - y.mean()) ** 2).sum())
+        return 1.0 - ss_res / ss_tot if ss_tot > 0 else 0.0
+
     # Reference-parity camelCase setters -------------------------------
     def setKernel(self, factory: Callable[[], Kernel]):
         self._kernel_factory = factory

@@ -66,3 +66,14 @@ def test_classifier_clone():
     assert c._max_newton_iter == 50 and c._sigma2 == 1e-2
     c2 = clone(c)
     assert c2.get_params()["max_newton_iter"] == 50
+
+
+def test_default_score_r2_and_cross_val_without_scoring():
+    rng = np.random.default_rng(2)
+    X = rng.uniform(size=(300, 2))
+    y = np.sin(3 * X.sum(-1)) + 0.05 * rng.normal(size=300)
+    est = _est()
+    est.fit(X, y)
+    assert est.score(X, y) > 0.9          # R^2 on train
+    scores = cross_val_score(_est(), X, y, cv=3)   # default scorer
+    assert (scores > 0.8).all(), scores
