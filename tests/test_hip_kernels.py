@@ -1,7 +1,6 @@
 """GPU numerics tests: every HIP kernel against the plain-PyTorch fp32/fp64
 reference of the same op (required test shape — see repo instructions)."""
 
-
 import numpy as np
 import pytest
 import torch
@@ -321,7 +320,6 @@ def test_force_lu_fallback_matches(dev, ext):
 def test_fused_laplace_newton_vs_torch(dev, ext):
     """The fused Newton loop must converge the latent f to the same point as
     the batched torch loop, and the dispatched GPC objective must match."""
-    import os
     from spark_gp_amd import ops
     from spark_gp_amd.kernels import (ARDRBFKernel, EyeKernel, Scalar,
                                       compile_kernel)
