@@ -23,8 +23,13 @@ from .active_set import (ActiveSetProvider, GreedilyOptimizingActiveSetProvider,
 from .kernels import (ARDRBFKernel, EyeKernel, Kernel, Matern32Kernel,
                       Matern52Kernel, RBFKernel, Scalar, SumOfKernels,
                       WhiteNoiseKernel)
+from .likelihoods import (Likelihood, LogisticLikelihood,
+                          PoissonLikelihood)
 from .models import (GaussianProcessClassificationModel,
-                     GaussianProcessClassifier, GaussianProcessRegression,
+                     GaussianProcessClassifier,
+                     GaussianProcessPoissonModel,
+                     GaussianProcessPoissonRegression,
+                     GaussianProcessRegression,
                      GaussianProcessRegressionModel, load_model, save_model)
 from .parallel import Comm, get_comm, init_from_env
 from .ppa import NotPositiveDefiniteError
@@ -36,6 +41,8 @@ __version__ = "0.1.0"
 __all__ = [
     "GaussianProcessRegression", "GaussianProcessRegressionModel",
     "GaussianProcessClassifier", "GaussianProcessClassificationModel",
+    "GaussianProcessPoissonRegression", "GaussianProcessPoissonModel",
+    "Likelihood", "LogisticLikelihood", "PoissonLikelihood",
     "Kernel", "RBFKernel", "ARDRBFKernel", "Matern32Kernel",
     "Matern52Kernel", "EyeKernel", "WhiteNoiseKernel",
     "SumOfKernels", "Scalar",

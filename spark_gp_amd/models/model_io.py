@@ -20,6 +20,7 @@ from ..kernels import (ARDRBFKernel, ConstantTimesKernel, EyeKernel, Kernel,
                        Matern32Kernel, Matern52Kernel, RBFKernel, SumOfKernels,
                        TrainableScalarTimesKernel)
 from .classification import GaussianProcessClassificationModel
+from .poisson import GaussianProcessPoissonModel
 from .predictor import GaussianProjectedProcessRawPredictor
 from .regression import GaussianProcessRegressionModel
 
@@ -92,9 +93,12 @@ def save_model(model, path: str) -> None:
     from safetensors.torch import save_file
     os.makedirs(path, exist_ok=True)
     raw: GaussianProjectedProcessRawPredictor = model.raw
-    kind = ("classification"
-            if isinstance(model, GaussianProcessClassificationModel)
-            else "regression")
+    if isinstance(model, GaussianProcessClassificationModel):
+        kind = "classification"
+    elif isinstance(model, GaussianProcessPoissonModel):
+        kind = "poisson"
+    else:
+        kind = "regression"
     save_file({
         "magic_vector": raw.magic_vector.cpu().contiguous(),
         "magic_matrix": raw.magic_matrix.cpu().contiguous(),
@@ -117,4 +121,6 @@ def load_model(path: str, device: str = "cpu"):
         tensors["active_set"])
     if spec["kind"] == "classification":
         return GaussianProcessClassificationModel(raw)
+    if spec["kind"] == "poisson":
+        return GaussianProcessPoissonModel(raw)
     return GaussianProcessRegressionModel(raw)

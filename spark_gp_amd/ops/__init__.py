@@ -83,8 +83,13 @@ def nll_grad_generic(kernel: Kernel, theta: np.ndarray,
 
 def laplace_nll_grad(kernel: Kernel, theta: np.ndarray, X: torch.Tensor,
                      y: torch.Tensor, f: torch.Tensor, tol: float,
-                     max_newton_iter: int = 200):
-    if X.is_cuda and not _force_torch():
+                     max_newton_iter: int = 200, likelihood=None):
+    from ..likelihoods import LogisticLikelihood
+    logistic = likelihood is None or isinstance(likelihood,
+                                                LogisticLikelihood)
+    # the fused HIP Newton kernel implements the logistic link; other
+    # likelihoods run on the batched torch path
+    if X.is_cuda and logistic and not _force_torch():
         from ..kernels.compiled import compile_kernel
         cs = compile_kernel(kernel)
         hip = _load_hip()
@@ -106,7 +111,8 @@ def laplace_nll_grad(kernel: Kernel, theta: np.ndarray, X: torch.Tensor,
             return torch_backend.laplace_nll_grad(
                 kernel, theta, X, y, f, tol, max_newton_iter)
     return torch_backend.laplace_nll_grad(kernel, theta, X, y, f, tol,
-                                          max_newton_iter)
+                                          max_newton_iter,
+                                          likelihood=likelihood)
 
 
 def kmn_knm_and_kmny(kernel: Kernel, active: torch.Tensor,

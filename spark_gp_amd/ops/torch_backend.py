@@ -159,7 +159,8 @@ def nll_grad_generic(kernel: Kernel, theta: np.ndarray,
 def laplace_nll_grad(kernel: Kernel, theta: np.ndarray,
                      X: torch.Tensor, y: torch.Tensor, f: torch.Tensor,
                      tol: float, max_newton_iter: int = 200,
-                     newton: bool = True) -> Tuple[float, np.ndarray]:
+                     newton: bool = True,
+                     likelihood=None) -> Tuple[float, np.ndarray]:
     """Batched Newton iteration (R&W Algorithm 3.1 with step halving) +
     Algorithm 5.1 evidence/gradient, mirroring
     ``classification/GaussianProcessClassifier.scala:74-129``.
@@ -170,7 +171,13 @@ def laplace_nll_grad(kernel: Kernel, theta: np.ndarray,
 
     Experts converge at different Newton iteration counts: converged experts
     are masked out of subsequent iterations (per-expert convergence mask —
-    SURVEY.md hard part #5)."""
+    SURVEY.md hard part #5).
+
+    ``likelihood``: a ``spark_gp_amd.likelihoods.Likelihood`` — Algorithms
+    3.1/5.1 only touch the likelihood through log p and its first three
+    f-derivatives.  Default: the reference's logistic link."""
+    from ..likelihoods import LogisticLikelihood
+    lik = likelihood if likelihood is not None else LogisticLikelihood()
     kernel.set_hyperparameters(theta)
     K, dK = kernel.training_kernel_and_derivative(X)   # [E,k,k], [E,p,k,k]
     E, k = y.shape
@@ -184,28 +191,26 @@ def laplace_nll_grad(kernel: Kernel, theta: np.ndarray,
     # final per-expert state needed by the evidence computation
     L_out = torch.empty_like(K)
     a_out = torch.empty_like(f)
-    pi_out = torch.empty_like(f)
     sqw_out = torch.empty_like(f)
+    f_eval_out = f.clone()         # latent at the last Newton EVALUATION
 
     active = torch.ones(E, dtype=torch.bool, device=dev)
     if not newton:
         # f already converged (fused HIP Newton pre-pass): evaluate the
         # exit-state quantities of Algorithm 3.1 once at the current f
-        pi = torch.sigmoid(f)
-        w = pi * (1.0 - pi)
+        w = lik.w(f, y)
         sqw = torch.sqrt(w)
         B = eyek + sqw.unsqueeze(-1) * K * sqw.unsqueeze(-2)
         L_out = torch.linalg.cholesky(B)
-        grad_logp = y - pi
+        grad_logp = lik.d1(f, y)
         b = w * f + grad_logp
         Kb = (K @ b.unsqueeze(-1)).squeeze(-1)
         v = torch.cholesky_solve((sqw * Kb).unsqueeze(-1), L_out).squeeze(-1)
         a_out = b - sqw * v
-        pi_out, sqw_out = pi, sqw
+        sqw_out = sqw
         fc = (K @ a_out.unsqueeze(-1)).squeeze(-1)
         new_obj = (-0.5 * (a_out * fc).sum(-1).double()
-                   + torch.nn.functional.logsigmoid(
-                       (2.0 * y - 1.0) * fc).double().sum(-1))
+                   + lik.log_lik(fc, y).double().sum(-1))
         active = torch.zeros(E, dtype=torch.bool, device=dev)
 
     it = 0
@@ -213,12 +218,11 @@ def laplace_nll_grad(kernel: Kernel, theta: np.ndarray,
         it += 1
         idx = active.nonzero(as_tuple=True)[0]
         Ki, fi, yi = K[idx], f[idx], y[idx]
-        pi = torch.sigmoid(fi)
-        w = pi * (1.0 - pi)
+        w = lik.w(fi, yi)
         sqw = torch.sqrt(w)
         B = eyek + sqw.unsqueeze(-1) * Ki * sqw.unsqueeze(-2)
         Li = torch.linalg.cholesky(B)
-        grad_logp = yi - pi
+        grad_logp = lik.d1(fi, yi)
         b = w * fi + grad_logp
         Kb = (Ki @ b.unsqueeze(-1)).squeeze(-1)
         v = torch.cholesky_solve((sqw * Kb).unsqueeze(-1), Li).squeeze(-1)
@@ -226,14 +230,13 @@ def laplace_nll_grad(kernel: Kernel, theta: np.ndarray,
         si = step[idx].to(dt).unsqueeze(-1)
         f_cand = (1.0 - si) * fi + si * (Ki @ a.unsqueeze(-1)).squeeze(-1)
         obj_cand = (-0.5 * (a * f_cand).sum(-1).double()
-                    + torch.nn.functional.logsigmoid(
-                        (2.0 * yi - 1.0) * f_cand).double().sum(-1))
+                    + lik.log_lik(f_cand, yi).double().sum(-1))
 
         accept = obj_cand > old_obj[idx]
         # store the state computed at the current f for every active expert —
         # at loop exit it corresponds to the last iteration, as in the ref.
         L_out[idx], a_out[idx] = Li, a
-        pi_out[idx], sqw_out[idx] = pi, sqw
+        sqw_out[idx], f_eval_out[idx] = sqw, fi
 
         acc_idx = idx[accept]
         if acc_idx.numel():
@@ -247,19 +250,19 @@ def laplace_nll_grad(kernel: Kernel, theta: np.ndarray,
         active[idx] = still
 
     # ---- Algorithm 5.1: evidence and gradient --------------------------
-    L, a, pi, sqw = L_out, a_out, pi_out, sqw_out
+    L, a, sqw = L_out, a_out, sqw_out
     logZ = new_obj - torch.log(L.diagonal(dim1=-2, dim2=-1)).double().sum(-1)
 
     sqwD = torch.diag_embed(sqw)
     R = sqw.unsqueeze(-1) * torch.cholesky_solve(sqwD, L)         # [E,k,k]
     Cm = torch.linalg.solve_triangular(L, sqw.unsqueeze(-1) * K,
                                        upper=False)               # [E,k,k]
-    d3 = -(2.0 * pi - 1.0) * pi * pi * torch.exp(-f)
+    d3 = lik.d3_evidence(f_eval_out, f, y)
     diagK = K.diagonal(dim1=-2, dim2=-1)
     diagCtC = (Cm * Cm).sum(-2)
     s2 = -0.5 * (diagK - diagCtC) * d3                            # [E,k]
 
-    grad_logp = y - pi
+    grad_logp = lik.d1(f_eval_out, y)
     p = dK.shape[-3]
     grad = torch.zeros(p, dtype=torch.float64)
     KR = K @ R                                                    # [E,k,k]
