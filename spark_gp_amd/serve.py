@@ -25,6 +25,7 @@ import numpy as np
 
 from .models.classification import GaussianProcessClassificationModel
 from .models.model_io import load_model
+from .models.poisson import GaussianProcessPoissonModel
 
 
 def create_app(model_path: str, device: str = "cpu"):
@@ -35,6 +36,7 @@ def create_app(model_path: str, device: str = "cpu"):
 
     model = load_model(model_path, device=device)
     is_clf = isinstance(model, GaussianProcessClassificationModel)
+    is_poisson = isinstance(model, GaussianProcessPoissonModel)
     active = model.raw.active_set
     m, d = active.shape
 
@@ -46,8 +48,9 @@ def create_app(model_path: str, device: str = "cpu"):
 
     @app.get("/health")
     def health():
-        return {"status": "ok",
-                "kind": "classification" if is_clf else "regression",
+        kind = ("classification" if is_clf
+                else "poisson" if is_poisson else "regression")
+        return {"status": "ok", "kind": kind,
                 "m": int(m), "d": int(d), "device": device}
 
     @app.post("/predict")
@@ -62,6 +65,11 @@ def create_app(model_path: str, device: str = "cpu"):
             proba = model.predict_proba(X)[:, 1]
             return {"proba": proba.tolist(),
                     "label": (proba > 0.5).astype(int).tolist()}
+        if is_poisson:
+            mu, var = model.predict_latent(X)
+            return {"rate": model.predict(X).tolist(),
+                    "latent_mean": mu.tolist(),
+                    "latent_var": var.tolist()}
         if req.return_std:
             mean, std = model.predict(X, return_std=True)
             return {"mean": mean.tolist(), "std": std.tolist()}

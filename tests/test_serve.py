@@ -68,3 +68,22 @@ def test_serve_classification_roundtrip(tmp_path):
     r = client.post("/predict", json={"X": X[:20].tolist()}).json()
     assert (np.array(r["label"]) == y[:20]).mean() >= 0.9
     assert all(0.0 <= p <= 1.0 for p in r["proba"])
+
+
+def test_serve_poisson_roundtrip(tmp_path):
+    from fastapi.testclient import TestClient
+    from spark_gp_amd import GaussianProcessPoissonRegression, save_model
+    rng = np.random.default_rng(3)
+    X = rng.uniform(size=(400, 2))
+    y = rng.poisson(np.exp(1.0 + np.sin(3 * X.sum(-1)))).astype(np.float64)
+    model = (GaussianProcessPoissonRegression()
+             .setKernel(lambda: 1 * ARDRBFKernel(2))
+             .setDatasetSizeForExpert(50).setActiveSetSize(60)
+             .setSigma2(1e-2).setMaxIter(15).setSeed(0).setDevice("cpu")
+             .fit(X, y))
+    save_model(model, str(tmp_path))
+    client = TestClient(create_app(str(tmp_path)))
+    assert client.get("/health").json()["kind"] == "poisson"
+    r = client.post("/predict", json={"X": X[:10].tolist()}).json()
+    np.testing.assert_allclose(r["rate"], model.predict(X[:10]), rtol=1e-10)
+    assert all(v >= 0 for v in r["latent_var"])
