@@ -307,3 +307,30 @@ def test_rank_with_empty_shard_participates():
     X, _ = performance_benchmark_data(200, 3, seed=21, dtype=np.float64)
     yq = np.sin(X[:30].sum(-1) * 3.0)
     assert float(np.sqrt(np.mean((res[0] - yq) ** 2))) < 0.25
+
+
+# ---- distributed Poisson GP -------------------------------------------------
+
+def _w_poisson_fit(rank, world):
+    from spark_gp_amd import GaussianProcessPoissonRegression
+    from spark_gp_amd.kernels import ARDRBFKernel
+    rng = np.random.default_rng(31)
+    X = rng.uniform(size=(600, 2))
+    y = rng.poisson(np.exp(1.0 + np.sin(3 * X.sum(-1)))).astype(np.float64)
+    Xl, yl = _shard(X, y, rank, world)
+    model = (GaussianProcessPoissonRegression()
+             .setKernel(lambda: 1 * ARDRBFKernel(2))
+             .setDatasetSizeForExpert(50).setActiveSetSize(80)
+             .setSigma2(1e-2).setMaxIter(20).setSeed(1).setDevice("cpu")
+             .fit(Xl, yl))
+    return model.predict(X[:40])
+
+
+def test_distributed_poisson_agrees_across_ranks():
+    res = _spawn("_w_poisson_fit")
+    np.testing.assert_allclose(res[0], res[1], atol=1e-10)
+    rng = np.random.default_rng(31)
+    X = rng.uniform(size=(600, 2))
+    true_rate = np.exp(1.0 + np.sin(3 * X.sum(-1)))
+    rel = np.abs(res[0] - true_rate[:40]) / true_rate[:40]
+    assert np.median(rel) < 0.35, np.median(rel)
