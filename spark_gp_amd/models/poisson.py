@@ -90,10 +90,20 @@ class GaussianProcessPoissonModel:
             Xt = Xt.unsqueeze(0)
         return self.raw.predict(Xt, with_var=True)
 
-    def predict(self, X) -> np.ndarray:
-        """Posterior-expected count rate E[exp(f*)] = exp(mu + var/2)."""
+    def predict(self, X, return_std: bool = False):
+        """Posterior-expected count rate E[exp(f*)] = exp(mu + var/2).
+
+        With ``return_std``: the posterior-PREDICTIVE count std via the law
+        of total variance, Var[y*] = E[lambda] + Var[lambda] with lognormal
+        latent moments (Var[lambda] = (exp(var) - 1) exp(2 mu + var))."""
         mu, var = self._latent(X)
-        return torch.exp(mu + 0.5 * var.clamp_min(0.0)).cpu().numpy()
+        var = var.clamp_min(0.0)
+        rate = torch.exp(mu + 0.5 * var)
+        if not return_std:
+            return rate.cpu().numpy()
+        var_lam = torch.expm1(var) * torch.exp(2.0 * mu + var)
+        std = torch.sqrt(rate + var_lam)
+        return rate.cpu().numpy(), std.cpu().numpy()
 
     def predict_latent(self, X):
         """(mean, var) of the latent log-rate."""

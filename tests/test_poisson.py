@@ -104,3 +104,22 @@ def test_poisson_model_persistence(tmp_path):
     assert type(loaded).__name__ == "GaussianProcessPoissonModel"
     np.testing.assert_allclose(loaded.predict(X[:20]), model.predict(X[:20]),
                                rtol=1e-12)
+
+
+def test_poisson_predictive_std_covers_counts():
+    """Var[y*] = E[lambda] + Var[lambda]: the predictive std must be at
+    least sqrt(rate) (Poisson floor) and the empirical counts should fall
+    within ~3 predictive stds of the predicted rate."""
+    rng = np.random.default_rng(9)
+    X = rng.uniform(size=(2000, 2))
+    true_rate = np.exp(1.0 + np.sin(3 * X.sum(-1)))
+    y = rng.poisson(true_rate).astype(np.float64)
+    model = (GaussianProcessPoissonRegression()
+             .setKernel(lambda: 1 * ARDRBFKernel(2))
+             .setDatasetSizeForExpert(60).setActiveSetSize(100)
+             .setSigma2(1e-2).setMaxIter(20).setSeed(0).setDevice("cpu")
+             .fit(X, y))
+    rate, std = model.predict(X[:500], return_std=True)
+    assert (std >= np.sqrt(rate) - 1e-9).all()
+    cover = np.mean(np.abs(y[:500] - rate) <= 3.0 * std)
+    assert cover > 0.97, cover
