@@ -56,6 +56,18 @@ def parse_args():
 
 def main():
     args = parse_args()
+    if args.gpus > 1 and int(os.environ.get("WORLD_SIZE", "1")) <= 1:
+        # A bare `python bench.py --gpus N` must NOT silently measure one
+        # GPU: re-exec under torchrun so N ranks actually exist.  (The
+        # driver normally launches torchrun itself, in which case
+        # WORLD_SIZE is already set and this branch is skipped.)
+        import subprocess
+        import sys
+        cmd = [sys.executable, "-m", "torch.distributed.run",
+               "--standalone", "--local-addr", "127.0.0.1",
+               "--nnodes=1", f"--nproc-per-node={args.gpus}",
+               os.path.abspath(__file__)] + sys.argv[1:]
+        raise SystemExit(subprocess.call(cmd))
     from spark_gp_amd import GaussianProcessRegression, init_from_env, get_comm
     from spark_gp_amd.kernels import ARDRBFKernel
     from spark_gp_amd.data import shard_performance_benchmark_data
@@ -74,6 +86,10 @@ def main():
     init_from_env(device)
     comm = get_comm()
     rank, world = comm.rank, comm.world_size
+    if world != args.gpus:
+        raise RuntimeError(
+            f"bench.py --gpus {args.gpus} but the process group has "
+            f"world_size={world}; refusing to report a mislabeled result")
     if device.type == "cuda":
         torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", 0)))
 

@@ -164,6 +164,10 @@ class GaussianProcessParams:
         return self
 
     def setTol(self, v: float):
+        """Convergence tolerance for L-BFGS-B and the Laplace Newton loops.
+        Honored exactly on every path: on GPU the fused fp32 Newton kernel
+        converges to max(tol, 1e-5) and a float64 torch polish finishes any
+        tighter request from the warm latent."""
         self._tol = float(v)
         return self
 
@@ -275,17 +279,28 @@ class GaussianProcessCommons(GaussianProcessParams):
         """Active set -> PPA stats (C2 allreduce) -> magic quantities
         (``GaussianProcessCommons.scala:40-59``)."""
         kernel = self._get_kernel().set_hyperparameters(theta)
-        t0 = time.perf_counter()
+
+        def stage_clock():
+            # device sync at every stage boundary so the published per-stage
+            # attribution is true wall time, not enqueue time (async GPU
+            # work would otherwise be billed to whichever later stage first
+            # blocks on it)
+            if X.is_cuda:
+                torch.cuda.synchronize(X.device)
+            return time.perf_counter()
+
+        t0 = stage_clock()
         active = self._active_set_provider(
             self._active_set_size, X, y, kernel, theta, self._seed, comm)
-        instr.log_timing("active_set", time.perf_counter() - t0)
+        t1 = stage_clock()
+        instr.log_timing("active_set", t1 - t0)
 
-        t0 = time.perf_counter()
         KK, Ky = accumulate_ppa_stats(kernel, active, X, y, comm,
                                       precision=self._ppa_precision)
-        instr.log_timing("ppa_accumulate", time.perf_counter() - t0)
+        t2 = stage_clock()
+        instr.log_timing("ppa_accumulate", t2 - t1)
 
-        t0 = time.perf_counter()
         mv, mm = magic_vector_matrix(kernel, KK, Ky, active)
-        instr.log_timing("magic_solve", time.perf_counter() - t0)
+        t3 = stage_clock()
+        instr.log_timing("magic_solve", t3 - t2)
         return GaussianProjectedProcessRawPredictor(mv, mm, kernel, active)

@@ -79,6 +79,11 @@ class GaussianProcessPoissonRegression(GaussianProcessCommons):
 
 
 class GaussianProcessPoissonModel:
+    # training clips the latent at PoissonLikelihood.fmax (=30); clamp the
+    # extrapolated PPA latent mean to the same ceiling so an
+    # out-of-distribution query cannot overflow exp() to inf rate/std
+    fmax = 30.0
+
     def __init__(self, raw: GaussianProjectedProcessRawPredictor):
         self.raw = raw
         self._instr: Optional[Instrumentation] = None
@@ -97,6 +102,7 @@ class GaussianProcessPoissonModel:
         of total variance, Var[y*] = E[lambda] + Var[lambda] with lognormal
         latent moments (Var[lambda] = (exp(var) - 1) exp(2 mu + var))."""
         mu, var = self._latent(X)
+        mu = mu.clamp(max=self.fmax)
         var = var.clamp_min(0.0)
         rate = torch.exp(mu + 0.5 * var)
         if not return_std:

@@ -102,12 +102,15 @@ def laplace_nll_grad(kernel: Kernel, theta: np.ndarray, X: torch.Tensor,
             # with the reference's exact semantics.
             n_bad = hip.laplace_newton(cs, theta, X, y, f, tol,
                                        max_newton_iter)
-            if n_bad == 0:
+            if n_bad == 0 and tol >= hip.LAPLACE_MIN_TOL:
                 # Algorithm 5.1 at the converged f via the contraction
                 # form — no [E, p, k, k] derivative tensor
                 return torch_backend.laplace_evidence_compiled(
                     cs, theta, X, y, f)
-            # some experts fell back: torch Newton finishes them warm
+            # Some experts fell back, or the user asked for a tol tighter
+            # than the fp32 kernel's clamp (LAPLACE_MIN_TOL): the torch
+            # Newton polish continues from the warm f with the exact
+            # requested tolerance (a few cheap iterations).
             return torch_backend.laplace_nll_grad(
                 kernel, theta, X, y, f, tol, max_newton_iter)
     return torch_backend.laplace_nll_grad(kernel, theta, X, y, f, tol,

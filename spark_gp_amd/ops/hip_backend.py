@@ -91,6 +91,12 @@ def nll_grad_compiled(cs: CompiledKernel, theta: np.ndarray,
 # Laplace Newton pre-pass (GPC)
 # ---------------------------------------------------------------------------
 
+# The fused fp32 Newton loop cannot resolve objective changes below the
+# fp32 noise floor; tolerances tighter than this are finished by a torch
+# Newton polish from the warm latent (see ops.__init__.laplace_nll_grad).
+LAPLACE_MIN_TOL = 1e-5
+
+
 def supports_laplace(cs: CompiledKernel, X: torch.Tensor) -> bool:
     if cs is None or cs.base not in ("ard", "rbf"):
         return False
@@ -116,7 +122,7 @@ def laplace_newton(cs: CompiledKernel, theta: np.ndarray, X: torch.Tensor,
     # convergence with reference semantics, so cap the fp32 iterations (the
     # fp32 objective noise floor can sit above a tight tol) and loosen tol
     # to the fp32-representable level.
-    eff_tol = max(float(tol), 1e-5)
+    eff_tol = max(float(tol), LAPLACE_MIN_TOL)
     psi, sll, iters, bad = ext.fused_laplace_newton(
         X, y.to(torch.float32), f, scale, float(C), float(nu), eff_tol,
         min(int(max_newton), 40))

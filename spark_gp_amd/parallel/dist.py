@@ -138,7 +138,21 @@ def init_from_env(device: Optional[torch.device] = None) -> Comm:
     """Initialize torch.distributed from torchrun env vars if present.
 
     Uses the nccl (=RCCL on ROCm) backend when the target device is CUDA,
-    gloo otherwise.  Safe to call when WORLD_SIZE is absent (no-op)."""
+    gloo otherwise.  Safe to call when WORLD_SIZE is absent (no-op).
+
+    Failure semantics (SURVEY.md §5 failure-detection row): fail fast, no
+    elastic recovery — a rank that desyncs or dies must take the job down
+    instead of hanging the collective forever.  Concretely:
+
+    * every collective carries a timeout (default 600 s, override with
+      ``SPARK_GP_AMD_COMM_TIMEOUT_S``) enforced by the RCCL watchdog thread;
+    * ``TORCH_NCCL_ASYNC_ERROR_HANDLING=1`` (set here unless the user chose
+      a value) makes the watchdog ABORT the process on a timed-out or
+      errored collective rather than logging and hanging;
+    * recovery is restart-the-job, mirroring the reference's
+      delegate-to-Spark stance (there is no partial-world continue: the BCM
+      objective is a fixed-order sum over all ranks)."""
+    import datetime
     import os
     if dist is None or dist.is_initialized():
         return Comm()
@@ -148,6 +162,9 @@ def init_from_env(device: Optional[torch.device] = None) -> Comm:
         (device is None and torch.cuda.is_available())
     backend = "nccl" if use_cuda else "gloo"
     if use_cuda:
+        os.environ.setdefault("TORCH_NCCL_ASYNC_ERROR_HANDLING", "1")
         torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", 0)))
-    dist.init_process_group(backend=backend)
+    timeout = datetime.timedelta(
+        seconds=float(os.environ.get("SPARK_GP_AMD_COMM_TIMEOUT_S", "600")))
+    dist.init_process_group(backend=backend, timeout=timeout)
     return Comm()
