@@ -24,7 +24,7 @@ from .kernels import (ARDRBFKernel, EyeKernel, Kernel, Matern32Kernel,
                       Matern52Kernel, RBFKernel, Scalar, SumOfKernels,
                       WhiteNoiseKernel)
 from .likelihoods import (Likelihood, LogisticLikelihood,
-                          PoissonLikelihood)
+                          PoissonLikelihood, ProbitLikelihood)
 from .models import (GaussianProcessClassificationModel,
                      GaussianProcessClassifier,
                      GaussianProcessPoissonModel,
@@ -43,6 +43,7 @@ __all__ = [
     "GaussianProcessClassifier", "GaussianProcessClassificationModel",
     "GaussianProcessPoissonRegression", "GaussianProcessPoissonModel",
     "Likelihood", "LogisticLikelihood", "PoissonLikelihood",
+    "ProbitLikelihood",
     "Kernel", "RBFKernel", "ARDRBFKernel", "Matern32Kernel",
     "Matern52Kernel", "EyeKernel", "WhiteNoiseKernel",
     "SumOfKernels", "Scalar",

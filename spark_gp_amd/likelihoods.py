@@ -82,6 +82,47 @@ class LogisticLikelihood(Likelihood):
                                                device=y.device)).all())
 
 
+class ProbitLikelihood(Likelihood):
+    """Bernoulli with probit link, y in {0, 1}:  p(y=1|f) = Phi(f).
+
+    With t = 2y - 1 and z = t f (R&W §3.9 probit expressions):
+      log p = log Phi(z)
+      d1    = t h(z),            h(z) = phi(z)/Phi(z)  (inverse Mills ratio)
+      W     = h(z) (h(z) + z)        in (0, 1), log-concave
+      dW/df = t [ h + (z + 2h) h' ],  h' = -(z h + h^2)
+    Everything is computed through ``log_ndtr`` so deep tails (|f| large)
+    never underflow Phi."""
+
+    _LOG_SQRT_2PI = 0.9189385332046727    # log sqrt(2 pi)
+
+    def _tzh(self, f, y):
+        t = 2.0 * y - 1.0
+        z = t * f
+        log_phi = -0.5 * z * z - self._LOG_SQRT_2PI
+        h = torch.exp(log_phi - torch.special.log_ndtr(z))
+        return t, z, h
+
+    def log_lik(self, f, y):
+        return torch.special.log_ndtr((2.0 * y - 1.0) * f)
+
+    def d1(self, f, y):
+        t, _, h = self._tzh(f, y)
+        return t * h
+
+    def w(self, f, y):
+        _, z, h = self._tzh(f, y)
+        return h * (h + z)
+
+    def d3(self, f, y):
+        t, z, h = self._tzh(f, y)
+        hp = -(z * h + h * h)
+        return t * (h + (z + 2.0 * h) * hp)
+
+    def validate_targets(self, y):
+        return bool(torch.isin(y, torch.tensor([0.0, 1.0], dtype=y.dtype,
+                                               device=y.device)).all())
+
+
 class PoissonLikelihood(Likelihood):
     """Poisson counts with log link: y | f ~ Poisson(exp(f)), y in {0,1,2,..}.
 

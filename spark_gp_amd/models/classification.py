@@ -24,15 +24,32 @@ from .predictor import GaussianProjectedProcessRawPredictor
 
 class GaussianProcessClassifier(GaussianProcessCommons):
     _PARAMS = dict(GaussianProcessCommons._PARAMS,
-                   max_newton_iter="_max_newton_iter")
+                   max_newton_iter="_max_newton_iter",
+                   link="_link")
 
     def __init__(self, **params):
         self._max_newton_iter = 200
+        self._link = "logistic"
         super().__init__(**params)
 
     def setMaxNewtonIter(self, v: int):
         self._max_newton_iter = int(v)
         return self
+
+    def setLink(self, link: str):
+        """Response link: 'logistic' (the reference's sigmoid link,
+        default) or 'probit' (Phi link; BASELINE config 3).  Algorithms
+        3.1/5.1 only touch the likelihood through log p and its first
+        three f-derivatives (see ``likelihoods.py``)."""
+        if link not in ("logistic", "probit"):
+            raise ValueError("link must be 'logistic' or 'probit'")
+        self._link = link
+        return self
+
+    def _resolve_likelihood(self):
+        from ..likelihoods import LogisticLikelihood, ProbitLikelihood
+        return (LogisticLikelihood() if self._link == "logistic"
+                else ProbitLikelihood())
 
     def fit(self, X, y) -> "GaussianProcessClassificationModel":
         instr = Instrumentation("GaussianProcessClassifier")
@@ -49,6 +66,7 @@ class GaussianProcessClassifier(GaussianProcessCommons):
 
         groups = group_experts(Xt, yt, self._dataset_size_for_expert)
         kernel = self._get_kernel()
+        lik = self._resolve_likelihood()
         # latent f per expert, zero-initialized, warm-started across evals
         fs = [torch.zeros_like(yg) for _, _, yg in groups]
 
@@ -57,7 +75,7 @@ class GaussianProcessClassifier(GaussianProcessCommons):
             for (idx, Xg, yg), fg in zip(groups, fs):
                 nll, grad = ops.laplace_nll_grad(
                     kernel, theta, Xg, yg, fg, self._tol,
-                    self._max_newton_iter)
+                    self._max_newton_iter, likelihood=lik)
                 nll_total += nll
                 grad_total += grad
             return nll_total, grad_total
@@ -72,7 +90,7 @@ class GaussianProcessClassifier(GaussianProcessCommons):
             f_flat[idx] = fg.reshape(-1)
         raw = self._produce_predictor(instr, comm, Xt, f_flat, theta)
         instr.log_success()
-        model = GaussianProcessClassificationModel(raw)
+        model = GaussianProcessClassificationModel(raw, link=self._link)
         model._instr = instr
         self.model_ = model           # sklearn-interop handle
         return model
@@ -81,9 +99,21 @@ class GaussianProcessClassifier(GaussianProcessCommons):
 class GaussianProcessClassificationModel:
     num_classes = 2
 
-    def __init__(self, raw: GaussianProjectedProcessRawPredictor):
+    def __init__(self, raw: GaussianProjectedProcessRawPredictor,
+                 link: str = "logistic"):
         self.raw = raw
+        self.link = link
         self._instr: Optional[Instrumentation] = None
+
+    def _squash(self, f: torch.Tensor) -> torch.Tensor:
+        return (torch.sigmoid(f) if self.link == "logistic"
+                else torch.special.ndtr(f))
+
+    def _squash_np(self, z: np.ndarray) -> np.ndarray:
+        if self.link == "logistic":
+            return 1.0 / (1.0 + np.exp(-z))
+        from scipy.special import ndtr
+        return ndtr(z)
 
     def _latent(self, X) -> Tuple[torch.Tensor, torch.Tensor]:
         Xt = torch.as_tensor(X, dtype=self.raw.active_set.dtype,
@@ -113,9 +143,9 @@ class GaussianProcessClassificationModel:
             integ = Integrator(quadrature_points)
             p1 = integ.expected_of_function_of_normal_batch(
                 f.cpu().numpy(), var.clamp_min(0.0).cpu().numpy(),
-                lambda z: 1.0 / (1.0 + np.exp(-z)))
+                self._squash_np)
         else:
-            p1 = torch.sigmoid(f).cpu().numpy()
+            p1 = self._squash(f).cpu().numpy()
         return np.stack([1.0 - p1, p1], axis=-1)
 
     def predict(self, X) -> np.ndarray:

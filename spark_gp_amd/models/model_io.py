@@ -104,10 +104,13 @@ def save_model(model, path: str) -> None:
         "magic_matrix": raw.magic_matrix.cpu().contiguous(),
         "active_set": raw.active_set.cpu().contiguous(),
     }, os.path.join(path, "model.safetensors"))
+    spec = {"kind": kind,
+            "kernel": kernel_to_spec(raw.kernel),
+            "format_version": 1}
+    if kind == "classification":
+        spec["link"] = getattr(model, "link", "logistic")
     with open(os.path.join(path, "spec.json"), "w") as fh:
-        json.dump({"kind": kind,
-                   "kernel": kernel_to_spec(raw.kernel),
-                   "format_version": 1}, fh, indent=2)
+        json.dump(spec, fh, indent=2)
 
 
 def load_model(path: str, device: str = "cpu"):
@@ -120,7 +123,8 @@ def load_model(path: str, device: str = "cpu"):
         tensors["magic_vector"], tensors["magic_matrix"], kernel,
         tensors["active_set"])
     if spec["kind"] == "classification":
-        return GaussianProcessClassificationModel(raw)
+        return GaussianProcessClassificationModel(
+            raw, link=spec.get("link", "logistic"))
     if spec["kind"] == "poisson":
         return GaussianProcessPoissonModel(raw)
     return GaussianProcessRegressionModel(raw)

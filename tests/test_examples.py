@@ -4,6 +4,8 @@
 < 2.1 (``regression/examples/Airfoil.scala:24``), both on CPU world_size=1
 (BASELINE config 1)."""
 
+import os
+
 import numpy as np
 import pytest
 
@@ -54,3 +56,18 @@ def test_airfoil_gate_rmse():
     rmse_cv = cross_validate(factory, Xs, ys, num_folds=10, seed=13)
     print("Airfoil RMSE:", rmse_cv)
     assert rmse_cv < 2.1
+
+
+def test_mnist68_probit_reduced_scale():
+    """BASELINE config 3 shape (784-dim MNIST stand-in, probit link) at
+    reduced row/m so it runs on CPU; the full 11769xd784 m=1000 run happens
+    on the GPU (examples/mnist68.py defaults)."""
+    import importlib.util
+    spec = importlib.util.spec_from_file_location(
+        "mnist68", os.path.join(os.path.dirname(__file__), "..",
+                                "examples", "mnist68.py"))
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    acc = mod.main(["--rows", "1200", "--active-set", "150",
+                    "--max-iter", "20", "--device", "cpu"])
+    assert acc > 0.95, acc
