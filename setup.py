@@ -23,6 +23,7 @@ ext = CUDAExtension(
         "spark_gp_amd/ops/csrc/expert_nll.hip",
         "spark_gp_amd/ops/csrc/cross_syrk.hip",
         "spark_gp_amd/ops/csrc/laplace.hip",
+        "spark_gp_amd/ops/csrc/big_chol.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3"],

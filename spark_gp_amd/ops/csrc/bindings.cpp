@@ -25,6 +25,16 @@ extern "C" hipError_t launch_colsum_gemv(const void* Kc, const float* y,
                                          int c, int m, double* Ky,
                                          hipStream_t stream);
 
+extern "C" hipError_t launch_dpotrf_diag(double* A, long m, int jb,
+                                         double* Vout, int* bad,
+                                         hipStream_t stream);
+
+extern "C" hipError_t launch_dgemm64(int ta, int tb, int sub, int syrk,
+                                     const double* A, const double* B,
+                                     double* C, int M, int N, int K,
+                                     long lda, long ldb, long ldc,
+                                     hipStream_t stream);
+
 extern "C" hipError_t launch_fused_laplace_newton(
     const float* X, const float* y, float* f, const float* scale, float amp,
     float noise, int E, int k, int d, double tol, int max_newton,
@@ -226,7 +236,112 @@ bool fused_laplace_newton_supported(int64_t k, int64_t d) {
   return bytes <= 160 * 1024;
 }
 
+// ---------------------------------------------------------------------------
+// K13: blocked fp64 Cholesky path (big_chol.hip) — host orchestration
+// ---------------------------------------------------------------------------
+
+// In-place blocked Cholesky of the padded [mp, mp] fp64 matrix A (lower
+// triangle; strict upper left untouched).  V [nb, 64, 64] receives the
+// explicit inverse of each diagonal block; bad (int32 [1], zeroed by the
+// caller) is set on a non-PD pivot.  mp must be a multiple of 64.
+void dpotrf64(torch::Tensor A, torch::Tensor V, torch::Tensor bad) {
+  TORCH_CHECK(A.is_cuda() && A.dtype() == torch::kFloat64 && A.dim() == 2 &&
+              A.size(0) == A.size(1) && A.is_contiguous());
+  const long mp = A.size(0);
+  TORCH_CHECK(mp % 64 == 0, "dpotrf64 requires 64-padded matrices");
+  const long nb = mp / 64;
+  TORCH_CHECK(V.is_cuda() && V.dtype() == torch::kFloat64 &&
+              V.numel() == nb * 64 * 64 && V.is_contiguous());
+  TORCH_CHECK(bad.is_cuda() && bad.dtype() == torch::kInt32);
+  auto stream = current_stream();
+  double* a = A.data_ptr<double>();
+  double* v = V.data_ptr<double>();
+  int* bd = bad.data_ptr<int>();
+  for (long J = 0; J < nb; ++J) {
+    const long jb = J * 64;
+    check_hip(launch_dpotrf_diag(a, mp, (int)jb, v + J * 4096, bd, stream),
+              "dpotrf_diag");
+    const long nr = mp - jb - 64;
+    if (nr > 0) {
+      double* A21 = a + (size_t)(jb + 64) * mp + jb;
+      // panel solve L21 = A21 V_J^T (in place)
+      check_hip(launch_dgemm64(0, 1, 0, 0, A21, v + J * 4096, A21,
+                               (int)nr, 64, 64, mp, 64, mp, stream),
+                "dtrsm_panel");
+      // trailing A22 -= L21 L21^T (lower tiles only)
+      double* A22 = a + (size_t)(jb + 64) * mp + jb + 64;
+      check_hip(launch_dgemm64(0, 1, 1, 1, A21, A21, A22,
+                               (int)nr, (int)nr, 64, mp, mp, mp, stream),
+                "dsyrk_trailing");
+    }
+  }
+}
+
+// Solve (L L^T) X = B in place on B [mp, r] given the factored A (lower L)
+// and the diagonal-block inverses V from dpotrf64.
+void dchol_solve64(torch::Tensor A, torch::Tensor V, torch::Tensor B) {
+  TORCH_CHECK(A.is_cuda() && A.dtype() == torch::kFloat64 &&
+              A.is_contiguous() && A.dim() == 2);
+  TORCH_CHECK(B.is_cuda() && B.dtype() == torch::kFloat64 &&
+              B.is_contiguous() && B.dim() == 2 && B.size(0) == A.size(0));
+  const long mp = A.size(0), r = B.size(1);
+  TORCH_CHECK(mp % 64 == 0);
+  const long nb = mp / 64;
+  TORCH_CHECK(V.numel() == nb * 64 * 64);
+  auto stream = current_stream();
+  double* a = A.data_ptr<double>();
+  double* v = V.data_ptr<double>();
+  double* b = B.data_ptr<double>();
+  for (long I = 0; I < nb; ++I) {            // forward: L Y = B
+    const long jb = I * 64;
+    check_hip(launch_dgemm64(0, 0, 0, 0, v + I * 4096, b + jb * r,
+                             b + jb * r, 64, (int)r, 64, 64, r, r, stream),
+              "fwd_diag");
+    const long nr = mp - jb - 64;
+    if (nr > 0)
+      check_hip(launch_dgemm64(0, 0, 1, 0, a + (size_t)(jb + 64) * mp + jb,
+                               b + jb * r, b + (jb + 64) * r,
+                               (int)nr, (int)r, 64, mp, r, r, stream),
+                "fwd_update");
+  }
+  for (long I = nb - 1; I >= 0; --I) {       // backward: L^T X = Y
+    const long jb = I * 64;
+    check_hip(launch_dgemm64(1, 0, 0, 0, v + I * 4096, b + jb * r,
+                             b + jb * r, 64, (int)r, 64, 64, r, r, stream),
+              "bwd_diag");
+    if (jb > 0)
+      check_hip(launch_dgemm64(1, 0, 1, 0, a + (size_t)jb * mp, b + jb * r,
+                               b, (int)jb, (int)r, 64, mp, r, r, stream),
+                "bwd_update");
+  }
+}
+
+// Generic fp64 MFMA GEMM (layout verification + utility): C = op(A) op(B).
+torch::Tensor dgemm64(torch::Tensor A, torch::Tensor B, bool ta, bool tb) {
+  TORCH_CHECK(A.is_cuda() && A.dtype() == torch::kFloat64 && A.dim() == 2);
+  TORCH_CHECK(B.is_cuda() && B.dtype() == torch::kFloat64 && B.dim() == 2);
+  auto Ac = A.contiguous();
+  auto Bc = B.contiguous();
+  const long M = ta ? A.size(1) : A.size(0);
+  const long K = ta ? A.size(0) : A.size(1);
+  const long Kb = tb ? B.size(1) : B.size(0);
+  const long N = tb ? B.size(0) : B.size(1);
+  TORCH_CHECK(K == Kb, "inner dims differ");
+  auto C = torch::empty({M, N}, A.options());
+  check_hip(launch_dgemm64(ta ? 1 : 0, tb ? 1 : 0, 0, 0,
+                           Ac.data_ptr<double>(), Bc.data_ptr<double>(),
+                           C.data_ptr<double>(), (int)M, (int)N, (int)K,
+                           A.size(1), B.size(1), N, current_stream()),
+            "dgemm64");
+  return C;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("dpotrf64", &dpotrf64,
+          "blocked fp64 Cholesky (K13), in place, 64-padded");
+  mod.def("dchol_solve64", &dchol_solve64,
+          "blocked fp64 triangular solves (L L^T) X = B, in place on B");
+  mod.def("dgemm64", &dgemm64, "fp64 MFMA GEMM (test/utility)");
   mod.def("fused_laplace_newton", &fused_laplace_newton,
           "per-expert Laplace Newton loop to convergence (CDNA4)");
   mod.def("fused_laplace_newton_supported", &fused_laplace_newton_supported);

@@ -194,6 +194,92 @@ def kmn_knm_and_kmny(kernel: Kernel, active: torch.Tensor,
     return KK.double(), Ky
 
 
+# ---------------------------------------------------------------------------
+# K13: hand-written blocked fp64 Cholesky path for the magic solves
+# ---------------------------------------------------------------------------
+# Replaces torch.linalg (rocSOLVER) in ppa.magic_vector_matrix on GPU:
+# big_chol.hip's one-workgroup 64x64 diagonal factors + MFMA-f64 GEMM
+# tiles for panel solve / trailing SYRK / blocked triangular solves.
+# Matrices are padded to a multiple of 64 with an identity block (its
+# factor and inverse are exact, so the un-padded region is unaffected).
+
+
+def _pad64_spd(M: torch.Tensor) -> torch.Tensor:
+    m = M.shape[0]
+    mp = (m + 63) & ~63
+    A = torch.zeros(mp, mp, dtype=torch.float64, device=M.device)
+    A[:m, :m] = M
+    if mp > m:
+        A.diagonal()[m:] = 1.0
+    return A
+
+
+def _dpotrf(Apad: torch.Tensor):
+    mp = Apad.shape[0]
+    V = torch.empty(mp // 64, 64, 64, dtype=torch.float64,
+                    device=Apad.device)
+    bad = torch.zeros(1, dtype=torch.int32, device=Apad.device)
+    ext.dpotrf64(Apad, V, bad)
+    return V, bad
+
+
+def chol_factor64(M: torch.Tensor, max_tries: int = 6):
+    """Blocked fp64 Cholesky with the escalating-jitter ladder of
+    ppa._chol_with_jitter (PD check = the factor's breakdown flag, not an
+    eigSym pass).  Returns (L_padded [mp,mp], diag-block inverses V)."""
+    from ..ppa import NotPositiveDefiniteError
+    Apad = _pad64_spd(M)
+    V, bad = _dpotrf(Apad)
+    if int(bad.item()) == 0:
+        return Apad, V
+    scale = float(M.diagonal().abs().mean())
+    eps = 1e-12
+    m = M.shape[0]
+    for _ in range(max_tries):
+        jit = eps * scale * torch.eye(m, dtype=M.dtype, device=M.device)
+        Apad = _pad64_spd(M + jit)
+        V, bad = _dpotrf(Apad)
+        if int(bad.item()) == 0:
+            return Apad, V
+        eps *= 100.0
+    raise NotPositiveDefiniteError()
+
+
+def chol_solve64(L: torch.Tensor, V: torch.Tensor, B: torch.Tensor,
+                 m: int) -> torch.Tensor:
+    """X = (L L^T)^{-1} B for B [m, r]; returns [m, r]."""
+    mp = L.shape[0]
+    Bp = torch.zeros(mp, B.shape[1], dtype=torch.float64, device=B.device)
+    Bp[:m] = B
+    ext.dchol_solve64(L, V, Bp)
+    return Bp[:m]
+
+
+def chol_inverse64(L: torch.Tensor, V: torch.Tensor, m: int) -> torch.Tensor:
+    mp = L.shape[0]
+    B = torch.eye(mp, dtype=torch.float64, device=L.device)
+    ext.dchol_solve64(L, V, B)
+    return B[:m, :m]
+
+
+def magic_vector_matrix(kernel: Kernel, KK: torch.Tensor, Ky: torch.Tensor,
+                        active: torch.Tensor):
+    """GPU magic quantities, entirely on the hand-written K13 kernels
+    (ProjectedGaussianProcessHelper.scala:49-60 semantics; fp64)."""
+    active64 = active.double()
+    Kmm = kernel.training_kernel(active64)
+    nu = kernel.white_noise_var()
+    PD = nu * Kmm + KK
+
+    Lp, Vp = chol_factor64(PD)
+    m = KK.shape[0]
+    mv = chol_solve64(Lp, Vp, Ky.unsqueeze(-1), m).squeeze(-1)
+    PDinv = chol_inverse64(Lp, Vp, m)
+    Lm, Vm = chol_factor64(Kmm)
+    Kmminv = chol_inverse64(Lm, Vm, m)
+    return mv, PDinv * nu - Kmminv
+
+
 def cross_kernel(kernel: Kernel, Xtest: torch.Tensor,
                  Xtrain: torch.Tensor) -> torch.Tensor:
     cs = compile_kernel(kernel)

@@ -49,7 +49,15 @@ def magic_vector_matrix(kernel: Kernel, KK: torch.Tensor, Ky: torch.Tensor,
 
     Computed redundantly on every rank (cheaper than broadcasting at m<=8192,
     C4 in SURVEY.md §2.5); all ranks hold identical KK/Ky after allreduce so
-    results are identical."""
+    results are identical.
+
+    On GPU the factorization/solves/inverses run on the hand-written K13
+    kernels (big_chol.hip: blocked fp64 Cholesky + MFMA-f64 GEMM tiles);
+    the torch path below is the CPU oracle."""
+    if KK.is_cuda and not ops._force_torch():
+        hip = ops._load_hip()
+        if ops._require_hip_or_fallback("magic_vector_matrix"):
+            return hip.magic_vector_matrix(kernel, KK, Ky, active)
     active64 = active.double()
     Kmm = kernel.training_kernel(active64)            # includes noise diag
     nu = kernel.white_noise_var()
