@@ -96,11 +96,11 @@ def _spawn_rccl(world, port):
     results = {}
     try:
         for _ in range(world):
-            rank, status, payload = out_q.get(timeout=600)
+            rank, status, payload = out_q.get(timeout=240)
             results[rank] = (status, payload)
     finally:
         for p in procs:
-            p.join(timeout=120)
+            p.join(timeout=60)
             if p.is_alive():
                 p.terminate()
     return results
