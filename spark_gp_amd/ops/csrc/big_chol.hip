@@ -322,7 +322,11 @@ dgemm64_kernel(const double* __restrict__ A, const double* __restrict__ B,
     for (int b = 0; b < 2; ++b)
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int gi = i0 + wr + a * 16 + kg * 4 + r;
+        // D map measured on gfx950 (scripts/mfma_f64_probe.hip +
+        // gpurun_out/mfma_probe.txt): D[row][col] at lane l reg r has
+        // col = l&15, row = (l>>4) + 4*r  (r strides by 4 — NOT the
+        // (l>>4)*4 + r grouping of the f32 16x16 shapes).
+        const int gi = i0 + wr + a * 16 + kg + 4 * r;
         const int gj = j0 + wc + b * 16 + l16;
         if (gi < M && gj < N && (!SYRK || gi >= gj)) {
           double* p = &C[(size_t)gi * ldc + gj];
@@ -368,6 +372,7 @@ extern "C" hipError_t launch_dgemm64(int ta, int tb, int sub, int syrk,
     case 7:  DG_CASE(false, true, true, true); break;      // NT sub syrk
     case 8:  DG_CASE(true, false, false, false); break;    // TN store
     case 10: DG_CASE(true, false, true, false); break;     // TN sub
+    case 12: DG_CASE(true, true, false, false); break;     // TT store (tests)
     default: return hipErrorInvalidValue;
   }
 #undef DG_CASE
