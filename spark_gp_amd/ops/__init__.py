@@ -95,6 +95,18 @@ def laplace_nll_grad(kernel: Kernel, theta: np.ndarray, X: torch.Tensor,
         hip = _load_hip()
         if (cs is not None
                 and _require_hip_or_fallback("laplace_nll_grad")
+                and tol >= hip.LAPLACE_MIN_TOL
+                and hip.supports_laplace_evidence(cs, X)):
+            # fully fused K10+K11: Newton AND the Algorithm 5.1 evidence
+            # in one launch; None -> fp32 breakdown, warm torch fallback
+            res = hip.laplace_evidence(cs, theta, X, y, f, tol,
+                                       max_newton_iter)
+            if res is not None:
+                return res
+            return torch_backend.laplace_nll_grad(
+                kernel, theta, X, y, f, tol, max_newton_iter)
+        if (cs is not None
+                and _require_hip_or_fallback("laplace_nll_grad")
                 and hip.supports_laplace(cs, X)):
             # fused Newton loop runs each expert to convergence on the GPU
             # (updates f in place); the torch pass below then converges in

@@ -25,6 +25,13 @@ extern "C" hipError_t launch_colsum_gemv(const void* Kc, const float* y,
                                          int c, int m, double* Ky,
                                          hipStream_t stream);
 
+extern "C" hipError_t launch_fused_laplace_evidence(
+    const float* X, const float* y, float* f, const float* scale, float amp,
+    float noise, int E, int k, int d, double tol, int max_newton,
+    double* out_psi, double* out_sumlogl, int* out_iters, int* out_bad,
+    double* out_logz, double* out_grad, hipStream_t stream,
+    size_t* lds_used);
+
 extern "C" hipError_t launch_dpotrf_diag(double* A, long m, int jb,
                                          double* Vout, int* bad,
                                          hipStream_t stream);
@@ -236,6 +243,56 @@ bool fused_laplace_newton_supported(int64_t k, int64_t d) {
   return bytes <= 160 * 1024;
 }
 
+// Fused Newton + Algorithm 5.1 evidence/gradient (K11).  Runs the Newton
+// loop to convergence (f updated in place), then the evidence pass in the
+// same launch.  Returns (logz[E] f64, grad[E, d+2] f64 — beta grads then
+// amp then noise, all d(logZ)/d(theta) un-negated —, iters[E] i32,
+// bad[E] i32).
+std::vector<torch::Tensor> fused_laplace_evidence(torch::Tensor X,
+                                                  torch::Tensor y,
+                                                  torch::Tensor f,
+                                                  torch::Tensor scale,
+                                                  double amp, double noise,
+                                                  double tol,
+                                                  int64_t max_newton) {
+  TORCH_CHECK(X.is_cuda() && X.dtype() == torch::kFloat32 && X.dim() == 3);
+  TORCH_CHECK(y.is_cuda() && y.dtype() == torch::kFloat32 && y.dim() == 2);
+  TORCH_CHECK(f.is_cuda() && f.dtype() == torch::kFloat32 && f.dim() == 2
+              && f.is_contiguous(), "f must be contiguous fp32");
+  auto Xc = X.contiguous();
+  auto yc = y.contiguous();
+  auto sc = scale.contiguous();
+  const int E = X.size(0), k = X.size(1), d = X.size(2);
+  auto opts64 = torch::TensorOptions().dtype(torch::kFloat64).device(X.device());
+  auto opts32i = torch::TensorOptions().dtype(torch::kInt32).device(X.device());
+  auto psi = torch::empty({E}, opts64);
+  auto sll = torch::empty({E}, opts64);
+  auto logz = torch::empty({E}, opts64);
+  auto grad = torch::empty({E, d + 2}, opts64);
+  auto iters = torch::empty({E}, opts32i);
+  auto bad = torch::empty({E}, opts32i);
+  size_t lds = 0;
+  check_hip(launch_fused_laplace_evidence(
+                Xc.data_ptr<float>(), yc.data_ptr<float>(),
+                f.data_ptr<float>(), sc.data_ptr<float>(), (float)amp,
+                (float)noise, E, k, d, tol, (int)max_newton,
+                psi.data_ptr<double>(), sll.data_ptr<double>(),
+                iters.data_ptr<int>(), bad.data_ptr<int>(),
+                logz.data_ptr<double>(), grad.data_ptr<double>(),
+                current_stream(), &lds),
+            "fused_laplace_evidence");
+  return {logz, grad, iters, bad};
+}
+
+bool fused_laplace_evidence_supported(int64_t k, int64_t d) {
+  if (k > 128 || d > k || k < 1) return false;
+  // EXACT mirror of lap_lds_bytes(k, d, ev=1) in laplace.hip
+  int64_t bytes = a16i(8 * 10) + 2 * a16i(4 * k * sa_of(k)) +
+                  a16i(4 * tsz_of(k)) + 8 * a16i(4 * k) + a16i(4 * d) + 16 +
+                  2 * a16i(4 * k) + a16i(4 * d) + a16i(8 * (d + 2));
+  return bytes <= 160 * 1024;
+}
+
 // ---------------------------------------------------------------------------
 // K13: blocked fp64 Cholesky path (big_chol.hip) — host orchestration
 // ---------------------------------------------------------------------------
@@ -345,6 +402,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("fused_laplace_newton", &fused_laplace_newton,
           "per-expert Laplace Newton loop to convergence (CDNA4)");
   mod.def("fused_laplace_newton_supported", &fused_laplace_newton_supported);
+  mod.def("fused_laplace_evidence", &fused_laplace_evidence,
+          "fused Newton + Algorithm 5.1 evidence/gradient (K11, CDNA4)");
+  mod.def("fused_laplace_evidence_supported",
+          &fused_laplace_evidence_supported);
   mod.def("fused_expert_nll", &fused_expert_nll,
           "fused per-expert BCM nll+gradient primitives (CDNA4)");
   mod.def("fused_expert_nll_profile", &fused_expert_nll_impl,

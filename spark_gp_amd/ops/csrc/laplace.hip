@@ -45,6 +45,11 @@ struct LapLds {
   float* t1;    // k
   float* t2;    // k
   float* s2;    // d
+  // evidence-mode (EV) extras
+  float* vv;    // k   v = y - pi
+  float* s2v;   // k   Algorithm 5.1 s2 vector
+  float* betav; // d   raw base scale (beta), signed
+  double* gout; // d+2 gradient accumulators (beta..., amp, noise)
   double* red;  // 8
   double* misc; // 2
   int* bad;     // 1
@@ -67,16 +72,23 @@ static __host__ __device__ inline size_t lap_a16(size_t n) {
   return (n + 15) & ~(size_t)15;
 }
 
-static __host__ __device__ inline size_t lap_lds_bytes(int k, int d) {
+static __host__ __device__ inline size_t lap_lds_bytes(int k, int d,
+                                                       int ev = 0) {
   size_t off = lap_a16(sizeof(double) * 10);
   off += 2 * lap_a16(sizeof(float) * (size_t)k * lap_sa(k));  // KB + A
   off += lap_a16(sizeof(float) * lap_tsz(k));
   off += 8 * lap_a16(sizeof(float) * k);
   off += lap_a16(sizeof(float) * d);
+  if (ev) {
+    off += 2 * lap_a16(sizeof(float) * k);       // vv + s2v
+    off += lap_a16(sizeof(float) * d);           // betav
+    off += lap_a16(sizeof(double) * (d + 2));    // gout
+  }
   off += 16;
   return off;
 }
 
+template <bool EV>
 __device__ inline LapLds lap_carve(char* base, int k, int d) {
   LapLds L;
   char* p = base;
@@ -94,6 +106,12 @@ __device__ inline LapLds lap_carve(char* base, int k, int d) {
   L.t1 = (float*)p;    p += lap_a16(sizeof(float) * k);
   L.t2 = (float*)p;    p += lap_a16(sizeof(float) * k);
   L.s2 = (float*)p;    p += lap_a16(sizeof(float) * d);
+  if (EV) {
+    L.vv = (float*)p;    p += lap_a16(sizeof(float) * k);
+    L.s2v = (float*)p;   p += lap_a16(sizeof(float) * k);
+    L.betav = (float*)p; p += lap_a16(sizeof(float) * d);
+    L.gout = (double*)p; p += lap_a16(sizeof(double) * (d + 2));
+  }
   L.bad = (int*)p;
   return L;
 }
@@ -104,21 +122,24 @@ __device__ inline double log_sigmoid(double v) {
   return -log1p(exp(-v));
 }
 
-extern "C" __global__ void __launch_bounds__(WG)
-fused_laplace_newton_kernel(const float* __restrict__ Xg,   // [E, k, d]
-                            const float* __restrict__ yg,   // [E, k]
-                            float* __restrict__ fg,         // [E, k] inout
-                            const float* __restrict__ scale,
-                            const float amp, const float noise,
-                            const int k, const int d,
-                            const double tol, const int max_newton,
-                            double* __restrict__ out_psi,      // [E]
-                            double* __restrict__ out_sumlogl,  // [E]
-                            int* __restrict__ out_iters,       // [E]
-                            int* __restrict__ out_bad) {       // [E]
+template <bool EV>
+__global__ void __launch_bounds__(WG)
+fused_laplace_kernel(const float* __restrict__ Xg,   // [E, k, d]
+                     const float* __restrict__ yg,   // [E, k]
+                     float* __restrict__ fg,         // [E, k] inout
+                     const float* __restrict__ scale,
+                     const float amp, const float noise,
+                     const int k, const int d,
+                     const double tol, const int max_newton,
+                     double* __restrict__ out_psi,      // [E]
+                     double* __restrict__ out_sumlogl,  // [E]
+                     int* __restrict__ out_iters,       // [E]
+                     int* __restrict__ out_bad,         // [E]
+                     double* __restrict__ out_logz,     // [E]     (EV)
+                     double* __restrict__ out_grad) {   // [E,d+2] (EV)
   extern __shared__ char lds_raw[];
   const int SA = (int)lap_sa(k);
-  LapLds S = lap_carve(lds_raw, k, d);
+  LapLds S = lap_carve<EV>(lds_raw, k, d);
   const int e = blockIdx.x;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -136,6 +157,7 @@ fused_laplace_newton_kernel(const float* __restrict__ Xg,   // [E, k, d]
   for (int j = tid; j < d; j += WG) {
     float s = scale[j];
     S.s2[j] = s * s;
+    if (EV) S.betav[j] = s;
   }
   if (tid == 0) *S.bad = 0;
   __syncthreads();
@@ -234,8 +256,300 @@ fused_laplace_newton_kernel(const float* __restrict__ Xg,   // [E, k, d]
     if (!(fabs(old_obj - new_obj) > tol && step > tol)) { ++it; break; }
   }
 
+  // ---- evidence tail (Algorithm 5.1, fused — K11) --------------------
+  // classification/GaussianProcessClassifier.scala:113-128, computed at
+  // the converged latent with the same exit-state semantics as
+  // torch_backend.laplace_evidence_compiled (f_eval == f_final).  The
+  // dK_i contractions never materialize [p, k, k]: the beta-gradient per
+  // column j chains  s3_j = Bm_j - K sqw (V^T V (sqw Bm_j))  through the
+  // triangular V from the final factor, 8 columns per pass through T.
+  double logZ = 0.0;
+  if (EV && !*S.bad) {
+    // exit-state recompute at the converged f
+    for (int i = tid; i < k; i += WG) {
+      const float fi = S.fb[i];
+      const float p = 1.f / (1.f + __expf(-fi));
+      const float w = p * (1.f - p);
+      S.pi[i] = p;
+      S.sqw[i] = sqrtf(w);
+      S.vv[i] = S.yb[i] - p;
+      S.bv[i] = w * fi + S.vv[i];
+    }
+    if (tid == 0) S.misc[0] = 0.0;
+    __syncthreads();
+    {
+      const int nlow = k * (k + 1) / 2;
+      for (int f = tid; f < nlow; f += WG) {
+        int a, b;
+        tri_decode(f, a, b);
+        S.A[(size_t)a * SA + b] = (a == b ? 1.f : 0.f) +
+            S.sqw[a] * S.KB[(size_t)a * SA + b] * S.sqw[b];
+      }
+    }
+    __syncthreads();
+    chol_invert_lower(S.A, S.T, k, SA, tid, lane, S.bad, S.misc);
+  }
+  if (EV && !*S.bad) {
+    // a = b - sqw V^T V (sqw (K b));  fc = K a;  psi;  logZ
+    for (int i = tid; i < k; i += WG)
+      S.t1[i] = dotv(S.KB + (size_t)i * SA, S.bv, 0, k) * S.sqw[i];
+    __syncthreads();
+    for (int i = tid; i < k; i += WG)
+      S.t2[i] = dotv(S.A + (size_t)i * SA, S.t1, 0, i + 1);
+    __syncthreads();
+    for (int a = tid; a < k; a += WG)
+      S.t1[a] = dotm(S.t2, S.A + a, SA, a, k);
+    __syncthreads();
+    for (int i = tid; i < k; i += WG)
+      S.av[i] = S.bv[i] - S.sqw[i] * S.t1[i];
+    __syncthreads();
+    double part = 0.0;
+    for (int i = tid; i < k; i += WG) {
+      const float fc = dotv(S.KB + (size_t)i * SA, S.av, 0, k);
+      part += -0.5 * (double)S.av[i] * (double)fc
+              + log_sigmoid((double)((2.f * S.yb[i] - 1.f) * fc));
+    }
+    const double psi = block_sum(part, S.red, tid);
+    logZ = psi - 0.5 * S.misc[0];
+
+    // d3 into pi; zero diagKRK accumulators (t2) and gout
+    for (int i = tid; i < k; i += WG) {
+      const float p = S.pi[i];
+      S.pi[i] = -(2.f * p - 1.f) * p * p * __expf(-S.fb[i]);
+      S.t2[i] = 0.f;
+    }
+    for (int j = tid; j < d + 2; j += WG) S.gout[j] = 0.0;
+    __syncthreads();
+    // diagKRK_i = || V (sqw o K_:,i) ||^2  (V lower in A; K symmetric)
+    for (int f = tid; f < k * k; f += WG) {
+      const int i = f / k, j = f - i * k;
+      const float* vr = S.A + (size_t)j * SA;
+      const float* kr = S.KB + (size_t)i * SA;
+      float z0 = 0.f, z1 = 0.f;
+      int c = 0;
+      for (; c + 1 <= j; c += 2) {
+        z0 += vr[c] * S.sqw[c] * kr[c];
+        z1 += vr[c + 1] * S.sqw[c + 1] * kr[c + 1];
+      }
+      if (c <= j) z0 += vr[c] * S.sqw[c] * kr[c];
+      const float z = z0 + z1;
+      atomicAdd(&S.t2[i], z * z);
+    }
+    __syncthreads();
+    // s2 vector; then u0 = K v - nu v into t2
+    for (int i = tid; i < k; i += WG)
+      S.s2v[i] = -0.5f * (S.KB[(size_t)i * SA + i] - S.t2[i]) * S.pi[i];
+    __syncthreads();
+    for (int i = tid; i < k; i += WG)
+      S.t1[i] = dotv(S.KB + (size_t)i * SA, S.vv, 0, k) - noise * S.vv[i];
+    __syncthreads();
+    for (int i = tid; i < k; i += WG) S.t2[i] = S.t1[i];   // u0
+    __syncthreads();
+
+    // ---- column chunks: d beta-columns, then Ba (amp) and v (noise) --
+    float* wk1 = S.T;
+    float* wk2 = S.T + 8 * k;
+    float* bmb = S.T + 16 * k;
+    float* qb = S.T + 24 * k;
+    for (int j0 = 0; j0 < d + 2; j0 += 8) {
+      const int cl = min(8, d + 2 - j0);
+      for (int f = tid; f < k * 8; f += WG) {
+        const int a = f >> 3, cc = f & 7;
+        const int j = j0 + cc;
+        float x = 0.f, w1 = 0.f, w2 = 0.f;
+        if (cc < cl && j < d) {
+          x = Xe[(size_t)a * d + j];
+          w1 = x * S.vv[a];
+          w2 = x * x * S.vv[a];
+        }
+        qb[f] = x;
+        wk1[f] = w1;
+        wk2[f] = w2;
+      }
+      __syncthreads();
+      // U1 = Kc (X_j o v) into bmb; then U2 = Kc (X_j^2 o v) into wk1
+      for (int f = tid; f < k * 8; f += WG) {
+        const int a = f >> 3, cc = f & 7;
+        bmb[f] = dotm(S.KB + (size_t)a * SA, wk1 + cc, 8, 0, k)
+                 - noise * wk1[f];
+      }
+      __syncthreads();
+      for (int f = tid; f < k * 8; f += WG) {
+        const int a = f >> 3, cc = f & 7;
+        const float u2 = dotm(S.KB + (size_t)a * SA, wk2 + cc, 8, 0, k)
+                         - noise * wk2[f];
+        const int j = j0 + cc;
+        float bm;
+        if (j < d) {
+          const float x = qb[f];
+          bm = -2.f * S.betav[j]
+               * (x * x * S.t2[a] - 2.f * x * bmb[f] + u2);
+        } else if (j == d) {
+          bm = S.t2[a] / amp;            // Ba = Kb v = u0 / C
+        } else if (j == d + 1) {
+          bm = S.vv[a];                  // noise column: dK = I
+        } else {
+          bm = 0.f;
+        }
+        // in-place into bmb: this thread is the only reader of bmb[f]
+        // in this loop (the dotm reads wk2, not bmb)
+        bmb[f] = bm;
+      }
+      __syncthreads();
+      // chain: s3 = bm - K sqw (V^T V (sqw bm))
+      for (int f = tid; f < k * 8; f += WG) {
+        const int a = f >> 3;
+        qb[f] = S.sqw[a] * bmb[f];
+      }
+      __syncthreads();
+      for (int f = tid; f < k * 8; f += WG) {
+        const int j = f >> 3, cc = f & 7;
+        wk1[f] = dotm(S.A + (size_t)j * SA, qb + cc, 8, 0, j + 1);
+      }
+      __syncthreads();
+      for (int f = tid; f < k * 8; f += WG) {
+        const int a = f >> 3, cc = f & 7;
+        wk2[f] = dot4(S.A + a, SA, wk1 + cc, 8, a, k);
+      }
+      __syncthreads();
+      for (int f = tid; f < k * 8; f += WG) {
+        const int a = f >> 3;
+        qb[f] = S.sqw[a] * wk2[f];
+      }
+      __syncthreads();
+      for (int f = tid; f < k * 8; f += WG) {
+        const int a = f >> 3, cc = f & 7;
+        wk2[f] = bmb[f] - dotm(S.KB + (size_t)a * SA, qb + cc, 8, 0, k);
+      }
+      __syncthreads();
+      // gout[j] += sum_a s2v[a] * s3[a][cc]   (wave cc owns column cc)
+      {
+        const int ww = tid >> 6;
+        if (ww < cl) {
+          double g = 0.0;
+          for (int a = lane; a < k; a += 64)
+            g += (double)S.s2v[a] * (double)wk2[a * 8 + ww];
+          g = wave_sum(g);
+          if (lane == 0) S.gout[j0 + ww] += g;
+        }
+      }
+      __syncthreads();
+    }
+
+    // ---- lauum: Binv = V^T V in place (lower), then mirror -----------
+    {
+      const int nblk = (k + NB - 1) / NB;
+      for (int I = 0; I < nblk; ++I) {
+        const int ib = I * NB;
+        const int bs = min(NB, k - ib);
+        const int ncol = ib + bs;
+        for (int f = tid; f < bs * ncol; f += WG) {
+          const int r = f / ncol, j = f - r * ncol;
+          const int i = ib + r;
+          if (j > i) continue;
+          S.T[r * SA + j] = dot4(S.A + i, SA, S.A + j, SA, i, k);
+        }
+        __syncthreads();
+        for (int f = tid; f < bs * ncol; f += WG) {
+          const int r = f / ncol, j = f - r * ncol;
+          if (j > ib + r) continue;
+          S.A[(size_t)(ib + r) * SA + j] = S.T[r * SA + j];
+        }
+        __syncthreads();
+      }
+      const int nlow = k * (k + 1) / 2;
+      for (int f = tid; f < nlow; f += WG) {
+        int a, b;
+        tri_decode(f, a, b);
+        if (a != b) S.A[(size_t)b * SA + a] = S.A[(size_t)a * SA + b];
+      }
+      __syncthreads();
+    }
+
+    // ---- W phase: scalars + W0 = (a a^T - R) o Kb in place -----------
+    double trR_p = 0.0, sRKb_p = 0.0, aKba_p = 0.0, a2_p = 0.0;
+    {
+      const int nlow = k * (k + 1) / 2;
+      for (int f = tid; f < nlow; f += WG) {
+        int a, b;
+        tri_decode(f, a, b);
+        const float binv = S.A[(size_t)a * SA + b];
+        const float R = S.sqw[a] * S.sqw[b] * binv;
+        const float kb = (S.KB[(size_t)a * SA + b] - (a == b ? noise : 0.f))
+                         / amp;
+        const float g = S.av[a] * S.av[b] - R;
+        const float w = g * kb;
+        if (a == b) {
+          trR_p += (double)R;
+          sRKb_p += (double)R * kb;
+          aKba_p += (double)S.av[a] * S.av[b] * kb;
+          a2_p += (double)S.av[a] * S.av[a];
+        } else {
+          sRKb_p += 2.0 * (double)R * kb;
+          aKba_p += 2.0 * (double)S.av[a] * S.av[b] * kb;
+        }
+        S.A[(size_t)a * SA + b] = w;
+        if (a != b) S.A[(size_t)b * SA + a] = w;
+      }
+    }
+    const double trR = block_sum(trR_p, S.red, tid);
+    const double sRKb = block_sum(sRKb_p, S.red, tid);
+    const double aKba = block_sum(aKba_p, S.red, tid);
+    const double a2 = block_sum(a2_p, S.red, tid);
+    if (tid == 0) {
+      S.gout[d] += 0.5 * aKba - 0.5 * sRKb;        // s1 for amp
+      S.gout[d + 1] += 0.5 * a2 - 0.5 * trR;       // s1 for noise
+    }
+    // r_a = row sums of W0 into t1
+    for (int a = tid; a < k; a += WG) {
+      const float* wr = S.A + (size_t)a * SA;
+      float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
+      int b = 0;
+      for (; b + 3 < k; b += 4) {
+        s0 += wr[b]; s1 += wr[b + 1]; s2 += wr[b + 2]; s3 += wr[b + 3];
+      }
+      for (; b < k; ++b) s0 += wr[b];
+      S.t1[a] = (s0 + s1) + (s2 + s3);
+    }
+    __syncthreads();
+    // s1 for beta: per 8-col X chunks: WX = W0 X; s1_j = -b_j C (2t1-2t2)
+    for (int d0 = 0; d0 < d; d0 += 8) {
+      const int cl = min(8, d - d0);
+      for (int f = tid; f < k * 8; f += WG) {
+        const int a = f >> 3, cc = f & 7;
+        qb[f] = (cc < cl) ? Xe[(size_t)a * d + d0 + cc] : 0.f;
+      }
+      __syncthreads();
+      for (int f = tid; f < k * 8; f += WG) {
+        const int a = f >> 3, cc = f & 7;
+        wk1[f] = dotm(S.A + (size_t)a * SA, qb + cc, 8, 0, k);
+      }
+      __syncthreads();
+      {
+        const int ww = tid >> 6;
+        if (ww < cl) {
+          double g = 0.0;
+          for (int a = lane; a < k; a += 64) {
+            const float x = qb[a * 8 + ww];
+            g += 2.0 * (double)x
+                 * ((double)x * (double)S.t1[a] - (double)wk1[a * 8 + ww]);
+          }
+          g = wave_sum(g);
+          if (lane == 0)
+            S.gout[d0 + ww] += -(double)S.betav[d0 + ww] * (double)amp * g;
+        }
+      }
+      __syncthreads();
+    }
+  }
+
   // ---- outputs -------------------------------------------------------
   for (int i = tid; i < k; i += WG) fg[(size_t)e * k + i] = S.fb[i];
+  if (EV) {
+    if (tid == 0) out_logz[e] = logZ;
+    for (int j = tid; j < d + 2; j += WG)
+      out_grad[(size_t)e * (d + 2) + j] = (*S.bad) ? 0.0 : S.gout[j];
+  }
   if (tid == 0) {
     out_bad[e] = *S.bad;
     out_psi[e] = new_obj;
@@ -254,8 +568,26 @@ extern "C" hipError_t launch_fused_laplace_newton(
   if (lds_used) *lds_used = lds;
   if (lds > 160 * 1024 || k > 128 || d > k)
     return hipErrorInvalidConfiguration;
-  hipLaunchKernelGGL(fused_laplace_newton_kernel, dim3(E), dim3(WG), lds,
+  hipLaunchKernelGGL(fused_laplace_kernel<false>, dim3(E), dim3(WG), lds,
                      stream, X, y, f, scale, amp, noise, k, d, tol,
-                     max_newton, out_psi, out_sumlogl, out_iters, out_bad);
+                     max_newton, out_psi, out_sumlogl, out_iters, out_bad,
+                     nullptr, nullptr);
+  return hipGetLastError();
+}
+
+extern "C" hipError_t launch_fused_laplace_evidence(
+    const float* X, const float* y, float* f, const float* scale, float amp,
+    float noise, int E, int k, int d, double tol, int max_newton,
+    double* out_psi, double* out_sumlogl, int* out_iters, int* out_bad,
+    double* out_logz, double* out_grad, hipStream_t stream,
+    size_t* lds_used) {
+  size_t lds = lap_lds_bytes(k, d, 1);
+  if (lds_used) *lds_used = lds;
+  if (lds > 160 * 1024 || k > 128 || d > k)
+    return hipErrorInvalidConfiguration;
+  hipLaunchKernelGGL(fused_laplace_kernel<true>, dim3(E), dim3(WG), lds,
+                     stream, X, y, f, scale, amp, noise, k, d, tol,
+                     max_newton, out_psi, out_sumlogl, out_iters, out_bad,
+                     out_logz, out_grad);
   return hipGetLastError();
 }

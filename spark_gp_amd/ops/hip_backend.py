@@ -129,6 +129,57 @@ def laplace_newton(cs: CompiledKernel, theta: np.ndarray, X: torch.Tensor,
     return int((bad != 0).sum())
 
 
+def supports_laplace_evidence(cs: CompiledKernel, X: torch.Tensor) -> bool:
+    if cs is None or cs.base not in ("ard", "rbf"):
+        return False
+    if X.dtype != torch.float32 or X.dim() != 3:
+        return False
+    E, k, d = X.shape
+    return bool(ext.fused_laplace_evidence_supported(k, d))
+
+
+def laplace_evidence(cs: CompiledKernel, theta: np.ndarray, X: torch.Tensor,
+                     y: torch.Tensor, f: torch.Tensor, tol: float,
+                     max_newton: int):
+    """Fully fused K10+K11: Newton to convergence AND the Algorithm 5.1
+    evidence/gradient in one launch (laplace.hip evidence tail).  Updates
+    f in place.  Returns (nll, grad) with the torch-path sign convention,
+    or None when any expert's fp32 factor broke down (caller falls back
+    to the warm torch Newton)."""
+    C = cs.amp(theta)
+    nu = cs.noise(theta)
+    d = X.shape[-1]
+    scale = _scale_vector(cs, theta, d, X.device)
+    if not f.is_contiguous():
+        raise ValueError("latent f must be contiguous")
+    eff_tol = max(float(tol), LAPLACE_MIN_TOL)
+    logz, grad, iters, bad = ext.fused_laplace_evidence(
+        X, y.to(torch.float32), f, scale, float(C), float(nu), eff_tol,
+        min(int(max_newton), 40))
+    # one device->host transfer for everything
+    stats = torch.cat([logz.sum().reshape(1), bad.sum().double().reshape(1),
+                       grad.sum(0)]).cpu().numpy()
+    if int(stats[1]):
+        return None
+    logZ = float(stats[0])
+    g = stats[2:]
+    gb, gamp, gnoise = g[:d], float(g[d]), float(g[d + 1])
+    out = np.zeros(cs.p)
+    if cs.amp_idx is not None:
+        out[cs.amp_idx] = gamp
+    if cs.base == "ard":
+        out[cs.base_idx] = gb
+    else:
+        sigma = float(theta[cs.base_idx][0])
+        # the kernel differentiates w.r.t. the uniform scale beta =
+        # 1/(sqrt(2) sigma); chain rule: dbeta/dsigma = -1/(sqrt(2) s^2)
+        out[cs.base_idx.start] = (-float(gb.sum())
+                                  / (math.sqrt(2.0) * sigma * sigma))
+    for i in cs.noise_idx:
+        out[i] += gnoise
+    return float(-logZ), -out
+
+
 # ---------------------------------------------------------------------------
 # PPA path
 # ---------------------------------------------------------------------------

@@ -352,3 +352,47 @@ def test_fused_laplace_newton_vs_torch(dev, ext):
     assert nll1 == pytest.approx(nll2, rel=1e-3)
     np.testing.assert_allclose(grad1, grad2, rtol=2e-2,
                                atol=1e-3 * np.abs(grad2).max())
+
+
+@pytest.mark.parametrize("base", ["ard", "rbf"])
+def test_fused_laplace_evidence_vs_torch(dev, ext, base):
+    """K11: the fully fused Newton + Algorithm 5.1 evidence/gradient launch
+    against the fp64 torch oracle (same warm-start, same tol)."""
+    from spark_gp_amd import ops
+    from spark_gp_amd.kernels import (ARDRBFKernel, EyeKernel, RBFKernel,
+                                      Scalar, compile_kernel)
+    from spark_gp_amd.ops import hip_backend, torch_backend
+    E, k, d = 8, 100, 8
+    g = torch.Generator().manual_seed(11)
+    X = torch.rand(E, k, d, generator=g).to(dev)
+    y = (X.sum(-1) > d / 2).to(torch.float32)
+    if base == "ard":
+        kernel = 1 * ARDRBFKernel(d) + Scalar(1e-2).const * EyeKernel()
+        theta = np.concatenate([[1.0], np.full(d, 0.8)])
+    else:
+        kernel = 1 * RBFKernel(0.8) + Scalar(1e-2).const * EyeKernel()
+        theta = np.array([1.1, 0.7])
+    cs = compile_kernel(kernel)
+    tol = 1e-5                        # >= LAPLACE_MIN_TOL: fused path ok
+
+    assert hip_backend.supports_laplace_evidence(cs, X)
+    f_hip = torch.zeros(E, k, device=dev)
+    res = hip_backend.laplace_evidence(cs, theta, X, y, f_hip, tol, 200)
+    assert res is not None, "fp32 breakdown on a benign batch"
+    nll_hip, grad_hip = res
+
+    f_ref = torch.zeros(E, k, dtype=torch.float64)
+    nll_ref, grad_ref = torch_backend.laplace_nll_grad(
+        kernel, theta, X.double().cpu(), y.double().cpu(), f_ref, tol)
+    assert nll_hip == pytest.approx(nll_ref, rel=1e-3)
+    np.testing.assert_allclose(grad_hip, grad_ref, rtol=2e-2,
+                               atol=1e-3 * np.abs(grad_ref).max())
+    # converged latents agree
+    np.testing.assert_allclose(f_hip.cpu().numpy(), f_ref.numpy(),
+                               rtol=2e-3, atol=2e-3)
+
+    # the dispatcher must route to the fused path at this tol
+    f2 = torch.zeros(E, k, device=dev)
+    nll2, grad2 = ops.laplace_nll_grad(kernel, theta, X, y, f2, tol)
+    assert nll2 == pytest.approx(nll_hip, rel=1e-9)
+    np.testing.assert_allclose(grad2, grad_hip, rtol=1e-9)
