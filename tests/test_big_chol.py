@@ -101,11 +101,14 @@ def test_jitter_ladder_recovers_semidefinite():
 
 
 def test_magic_vector_matrix_hip_matches_cpu_oracle():
+    # sigma2 = 0.1 keeps cond(PD) ~ 1e7: the comparison then reflects the
+    # K13 implementation, not conditioning-amplified 1e-16 input noise
+    # (Kmm is rebuilt per device; sqdist GEMM order differs CPU vs GPU)
     import spark_gp_amd.ppa as ppa
     from spark_gp_amd.kernels import ARDRBFKernel, EyeKernel, Scalar
     torch.manual_seed(0)
-    m, d, n = 300, 4, 2000
-    kernel = 1 * ARDRBFKernel(d) + Scalar(1e-3).const * EyeKernel()
+    m, d, n = 200, 4, 1000
+    kernel = 1 * ARDRBFKernel(d) + Scalar(0.1).const * EyeKernel()
     kernel.set_hyperparameters(
         np.array([1.1] + [0.9] * d))
     X = torch.rand(n, d, dtype=torch.float64)
