@@ -30,12 +30,18 @@ def main():
     p.add_argument("--tol", type=float, default=1e-5)
     p.add_argument("--fits", type=int, default=3)
     p.add_argument("--warmup", type=int, default=1)
+    p.add_argument("--ppa-precision", type=str, default="fp64",
+                   choices=["fp64", "mixed"])
     args = p.parse_args()
 
     dev = "cuda" if torch.cuda.is_available() else "cpu"
     rng = np.random.default_rng(13)
     X = rng.random((args.rows, args.dim), dtype=np.float64)
-    y = (np.sin(4.0 * X.sum(-1)) > 0).astype(np.float64)
+    # learnable decision boundary: only 2 active dims at a lengthscale the
+    # ARD-RBF can represent with m=1000 (a sin(4*sum of ALL dims) target is
+    # label noise at d=16 and lets Newton exit immediately — it would
+    # understate the Laplace work)
+    y = (np.sin(4.0 * (X[:, 0] + X[:, 1])) > 0).astype(np.float64)
     Xt = torch.tensor(X, dtype=torch.float32, device=dev)
     yt = torch.tensor(y, dtype=torch.float32, device=dev)
 
@@ -45,6 +51,7 @@ def main():
                 .setDatasetSizeForExpert(args.expert_size)
                 .setActiveSetSize(args.active_set)
                 .setSigma2(1e-3)
+                .setPpaPrecision(args.ppa_precision)
                 .setTol(args.tol)
                 .setMaxIter(args.max_iter)
                 .setSeed(13)
