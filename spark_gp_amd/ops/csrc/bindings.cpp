@@ -120,8 +120,9 @@ inline int64_t tsz_of(int64_t k) {
 bool fused_expert_nll_supported(int64_t k, int64_t d) {
   if (k > 128 || d > 128 || k < 1) return false;
   // LDS budget: EXACT mirror of nll_lds_bytes2 in expert_nll.hip
+  const int64_t dp4 = (d + 4) & ~(int64_t)3;
   int64_t bytes = a16i(8 * 10) + a16i(4 * k * sa_of(k)) +
-                  a16i(4 * tsz_of(k)) + a16i(4 * k * (d + 1)) +
+                  a16i(4 * tsz_of(k)) + a16i(4 * k * dp4) +
                   4 * a16i(4 * k) + a16i(4 * d) + 16;
   return bytes <= 160 * 1024;
 }

@@ -82,12 +82,16 @@ static __host__ __device__ inline size_t a16(size_t n) {
   return (n + 15) & ~(size_t)15;
 }
 
+static __host__ __device__ inline size_t nll_dp4(int d) {
+  return (size_t)((d + 4) & ~3);     // X row stride: 16-B aligned rows
+}
+
 static __host__ __device__ inline size_t nll_lds_bytes2(int k, int d) {
   size_t off = 0;
   off += a16(sizeof(double) * 10);                     // red + misc
   off += a16(sizeof(float) * (size_t)k * nll_sa(k));   // A
   off += a16(sizeof(float) * nll_tsz(k));              // T
-  off += a16(sizeof(float) * (size_t)k * (d + 1));     // X
+  off += a16(sizeof(float) * (size_t)k * nll_dp4(d));  // X
   off += 4 * a16(sizeof(float) * k);                   // yb alpha tvec rrow
   off += a16(sizeof(float) * d);                       // s2
   off += 16;                                           // bad (+pad)
@@ -101,7 +105,7 @@ __device__ inline NllLds carve(char* base, int k, int d) {
   p += a16(sizeof(double) * 10);
   L.A = (float*)p;      p += a16(sizeof(float) * (size_t)k * nll_sa(k));
   L.T = (float*)p;      p += a16(sizeof(float) * nll_tsz(k));
-  L.X = (float*)p;      p += a16(sizeof(float) * (size_t)k * (d + 1));
+  L.X = (float*)p;      p += a16(sizeof(float) * (size_t)k * nll_dp4(d));
   L.yb = (float*)p;     p += a16(sizeof(float) * k);
   L.alpha = (float*)p;  p += a16(sizeof(float) * k);
   L.tvec = (float*)p;   p += a16(sizeof(float) * k);
@@ -130,7 +134,7 @@ fused_expert_nll_kernel(const float* __restrict__ Xg,
   PH(0);
   NllLds S = carve(lds_raw, k, d);
   const int SA = (int)nll_sa(k);
-  const int dp = d + 1;
+  const int dp = (int)nll_dp4(d);
   const int e = blockIdx.x;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -160,16 +164,26 @@ PH(1);
       tri_decode(f, a, b);
       const float* xa = S.X + a * dp;
       const float* xb = S.X + b * dp;
-      float q0 = 0.f, q1 = 0.f;
+      // float4 (ds_read_b128) over the 16-B-aligned X rows and s2
+      float4 q4 = {0.f, 0.f, 0.f, 0.f};
       int j = 0;
-      for (; j + 1 < d; j += 2) {
-        float t0 = xa[j] - xb[j];
-        float t1 = xa[j + 1] - xb[j + 1];
-        q0 += S.s2[j] * t0 * t0;
-        q1 += S.s2[j + 1] * t1 * t1;
+      for (; j + 3 < d; j += 4) {
+        const float4 va = *(const float4*)(xa + j);
+        const float4 vb = *(const float4*)(xb + j);
+        const float4 sv = *(const float4*)(S.s2 + j);
+        const float u0 = va.x - vb.x, u1 = va.y - vb.y;
+        const float u2 = va.z - vb.z, u3 = va.w - vb.w;
+        q4.x += sv.x * u0 * u0;
+        q4.y += sv.y * u1 * u1;
+        q4.z += sv.z * u2 * u2;
+        q4.w += sv.w * u3 * u3;
       }
-      if (j < d) { float t = xa[j] - xb[j]; q0 += S.s2[j] * t * t; }
-      const float kb = __expf(-(q0 + q1));
+      float q0 = (q4.x + q4.y) + (q4.z + q4.w);
+      for (; j < d; ++j) {
+        const float t = xa[j] - xb[j];
+        q0 += S.s2[j] * t * t;
+      }
+      const float kb = __expf(-q0);
       S.A[a * SA + b] = amp * kb + (a == b ? noise : 0.f);
       if (a != b) S.A[b * SA + a] = kb;        // Kb cache in the upper
     }
@@ -208,23 +222,41 @@ PH(3);
   const double yta = block_sum(part, S.red, tid);
 
 PH(5);
-    // ---- L: K^-1 = V^T V in place (lauum), ascending row blocks ------
-  // strictly j <= i so the Kb cache in the upper triangle survives
-  for (int I = 0; I < nblk; ++I) {
-    const int ib = I * NB;
-    const int bs = min(NB, k - ib);
-    const int ncol = ib + bs;
-    for (int f = tid; f < bs * ncol; f += WG) {
-      const int r = f / ncol, j = f - r * ncol;
-      const int i = ib + r;
-      if (j > i) continue;
-      S.T[r * SA + j] = dot4(S.A + i, SA, S.A + j, SA, i, k);
+  // ---- L: K^-1 = V^T V (lauum), register-accumulated ---------------
+  // Kinv_ij = sum_c V[c][i] V[c][j] (c >= i >= j).  The rows of V are
+  // staged TRANSPOSED in 32-column chunks through T (T[r][cc] = V[c0+cc]
+  // [r], row stride 36 floats = 16-B aligned), so every output pair is a
+  // contiguous float4 dot of two T rows instead of two SA-strided walks.
+  // Each thread owns fixed (i, j) pairs across chunks with register
+  // accumulators; A's lower is only overwritten after the last chunk
+  // (the Kb cache in the strict upper survives untouched).
+  {
+    constexpr int MAXP = (128 * 129 / 2 + WG - 1) / WG;   // 17 at k=128
+    float accs[MAXP];
+    const int nlow = k * (k + 1) / 2;
+#pragma unroll
+    for (int m = 0; m < MAXP; ++m) accs[m] = 0.f;
+    for (int c0 = 0; c0 < k; c0 += 32) {
+      const int cl = min(32, k - c0);
+      for (int f = tid; f < k * cl; f += WG) {
+        const int r = f / cl, cc = f - r * cl;
+        S.T[r * 36 + cc] = (r <= c0 + cc)
+                               ? S.A[(size_t)(c0 + cc) * SA + r] : 0.f;
+      }
+      __syncthreads();
+      int m = 0;
+      for (int f = tid; f < nlow; f += WG, ++m) {
+        int i, j;
+        tri_decode(f, i, j);
+        accs[m] += dotv(S.T + i * 36, S.T + j * 36, 0, cl);
+      }
+      __syncthreads();
     }
-    __syncthreads();
-    for (int f = tid; f < bs * ncol; f += WG) {
-      const int r = f / ncol, j = f - r * ncol;
-      if (j > ib + r) continue;
-      S.A[(size_t)(ib + r) * SA + j] = S.T[r * SA + j];
+    int m = 0;
+    for (int f = tid; f < nlow; f += WG, ++m) {
+      int i, j;
+      tri_decode(f, i, j);
+      S.A[(size_t)i * SA + j] = accs[m];
     }
     __syncthreads();
   }
@@ -268,24 +300,30 @@ PH(7);
   __syncthreads();
 
 PH(8);
-    // ---- H: WX = W0 @ X (d-halves through T); contraction ------------
+  // ---- H: contraction, X^T-staged ----------------------------------
+  // contr_j = 2 sum_a x_aj^2 r_a - 2 sum_a x_aj (W0 X)_aj.  X columns
+  // are staged transposed through T (T[j][a] = X[a][d0+j], 16-B-aligned
+  // 36-float rows) so (W0 X)_aj = dotv(W0 row a, T row j) is a float4
+  // dot of two contiguous rows; wave w owns columns j = w (mod 8), lanes
+  // stride the row index — no barriers inside a chunk.
   for (int d0 = 0; d0 < d; d0 += 32) {
     const int dl = min(32, d - d0);
     for (int f = tid; f < k * dl; f += WG) {
       const int a = f / dl, j = f - a * dl;
-      S.T[a * 36 + j] = dotm(S.A + (size_t)a * SA,
-                             S.X + d0 + j, dp, 0, k);
+      S.T[j * SA + a] = S.X[a * dp + d0 + j];
     }
     __syncthreads();
-    // contr_j = 2 sum_a x_aj^2 r_a - 2 sum_a x_aj WX_aj   (fp64)
-    for (int j = tid; j < dl; j += WG) {
+    const int ww = tid >> 6;
+    for (int j = ww; j < dl; j += 8) {
       double acc = 0.0;
-      for (int a = 0; a < k; ++a) {
-        const float x = S.X[a * dp + d0 + j];
-        acc += 2.0 * (double)x *
-               ((double)x * (double)S.rrow[a] - (double)S.T[a * 36 + j]);
+      for (int a = lane; a < k; a += 64) {
+        const float x = S.T[j * SA + a];
+        const float wx = dotv(S.A + (size_t)a * SA, S.T + j * SA, 0, k);
+        acc += 2.0 * (double)x * ((double)x * (double)S.rrow[a]
+                                  - (double)wx);
       }
-      out_contr[(size_t)e * d + d0 + j] = acc;
+      acc = wave_sum(acc);
+      if (lane == 0) out_contr[(size_t)e * d + d0 + j] = acc;
     }
     __syncthreads();
   }
