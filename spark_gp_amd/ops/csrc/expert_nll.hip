@@ -232,6 +232,9 @@ PH(5);
   // (the Kb cache in the strict upper survives untouched).
   {
     constexpr int MAXP = (128 * 129 / 2 + WG - 1) / WG;   // 17 at k=128
+    // fully-unrolled accumulator array (a runtime-indexed array would
+    // spill to scratch); (i, j) re-decoded per chunk — a few hundred
+    // cycles total, cheaper than 17 more registers of pair cache
     float accs[MAXP];
     const int nlow = k * (k + 1) / 2;
 #pragma unroll
@@ -244,16 +247,24 @@ PH(5);
                                ? S.A[(size_t)(c0 + cc) * SA + r] : 0.f;
       }
       __syncthreads();
-      int m = 0;
-      for (int f = tid; f < nlow; f += WG, ++m) {
+#pragma unroll
+      for (int m = 0; m < MAXP; ++m) {
+        const int f = tid + m * WG;
+        if (f >= nlow) break;
         int i, j;
         tri_decode(f, i, j);
-        accs[m] += dotv(S.T + i * 36, S.T + j * 36, 0, cl);
+        // contributions only from c >= i: clip the chunk range (the
+        // staged zeros make wider ranges correct but waste cycles)
+        const int lo = i > c0 ? i - c0 : 0;
+        if (lo < cl)
+          accs[m] += dotv(S.T + i * 36, S.T + j * 36, lo, cl);
       }
       __syncthreads();
     }
-    int m = 0;
-    for (int f = tid; f < nlow; f += WG, ++m) {
+#pragma unroll
+    for (int m = 0; m < MAXP; ++m) {
+      const int f = tid + m * WG;
+      if (f >= nlow) break;
       int i, j;
       tri_decode(f, i, j);
       S.A[(size_t)i * SA + j] = accs[m];
