@@ -194,8 +194,8 @@ PH(2);
   // ---- C/D: in-place blocked Cholesky + triangular inverse ---------
   // (shared machinery: linalg_lds.h)  A: lower K -> V = L^-1; upper Kb
   // cache untouched; log|K| into misc[0]; bad flag on fp32 breakdown.
-  chol_invert_lower(S.A, S.T, k, SA, tid, lane, S.bad, S.misc,
-                    out_clk ? out_clk + (size_t)e * 20 + 12 : nullptr);
+  chol_invert_lower_pipe(S.A, S.T, k, SA, tid, lane, S.bad, S.misc,
+                         out_clk ? out_clk + (size_t)e * 20 + 12 : nullptr);
   if (*S.bad) {
     if (tid == 0) {
       out_bad[e] = *S.bad;     // 1: indefinite, 2: non-finite iterate
@@ -222,52 +222,27 @@ PH(3);
   const double yta = block_sum(part, S.red, tid);
 
 PH(5);
-  // ---- L: K^-1 = V^T V (lauum), register-accumulated ---------------
-  // Kinv_ij = sum_c V[c][i] V[c][j] (c >= i >= j).  The rows of V are
-  // staged TRANSPOSED in 32-column chunks through T (T[r][cc] = V[c0+cc]
-  // [r], row stride 36 floats = 16-B aligned), so every output pair is a
-  // contiguous float4 dot of two T rows instead of two SA-strided walks.
-  // Each thread owns fixed (i, j) pairs across chunks with register
-  // accumulators; A's lower is only overwritten after the last chunk
-  // (the Kb cache in the strict upper survives untouched).
-  {
-    constexpr int MAXP = (128 * 129 / 2 + WG - 1) / WG;   // 17 at k=128
-    // fully-unrolled accumulator array (a runtime-indexed array would
-    // spill to scratch); (i, j) re-decoded per chunk — a few hundred
-    // cycles total, cheaper than 17 more registers of pair cache
-    float accs[MAXP];
-    const int nlow = k * (k + 1) / 2;
-#pragma unroll
-    for (int m = 0; m < MAXP; ++m) accs[m] = 0.f;
-    for (int c0 = 0; c0 < k; c0 += 32) {
-      const int cl = min(32, k - c0);
-      for (int f = tid; f < k * cl; f += WG) {
-        const int r = f / cl, cc = f - r * cl;
-        S.T[r * 36 + cc] = (r <= c0 + cc)
-                               ? S.A[(size_t)(c0 + cc) * SA + r] : 0.f;
-      }
-      __syncthreads();
-#pragma unroll
-      for (int m = 0; m < MAXP; ++m) {
-        const int f = tid + m * WG;
-        if (f >= nlow) break;
-        int i, j;
-        tri_decode(f, i, j);
-        // contributions only from c >= i: clip the chunk range (the
-        // staged zeros make wider ranges correct but waste cycles)
-        const int lo = i > c0 ? i - c0 : 0;
-        if (lo < cl)
-          accs[m] += dotv(S.T + i * 36, S.T + j * 36, lo, cl);
-      }
-      __syncthreads();
+  // ---- L: K^-1 = V^T V in place (lauum), ascending row blocks ------
+  // strictly j <= i so the Kb cache in the upper triangle survives.
+  // (A register-accumulated transposed-chunk variant measured SLOWER —
+  // 21.3 vs 18.5 us at k=100: the short per-chunk dots are overhead-
+  // bound and the unrolled accumulator array costs registers.  Negative
+  // result recorded in TODO.md; keep the strided dot4 form.)
+  for (int I = 0; I < nblk; ++I) {
+    const int ib = I * NB;
+    const int bs = min(NB, k - ib);
+    const int ncol = ib + bs;
+    for (int f = tid; f < bs * ncol; f += WG) {
+      const int r = f / ncol, j = f - r * ncol;
+      const int i = ib + r;
+      if (j > i) continue;
+      S.T[r * SA + j] = dot4(S.A + i, SA, S.A + j, SA, i, k);
     }
-#pragma unroll
-    for (int m = 0; m < MAXP; ++m) {
-      const int f = tid + m * WG;
-      if (f >= nlow) break;
-      int i, j;
-      tri_decode(f, i, j);
-      S.A[(size_t)i * SA + j] = accs[m];
+    __syncthreads();
+    for (int f = tid; f < bs * ncol; f += WG) {
+      const int r = f / ncol, j = f - r * ncol;
+      if (j > ib + r) continue;
+      S.A[(size_t)(ib + r) * SA + j] = S.T[r * SA + j];
     }
     __syncthreads();
   }

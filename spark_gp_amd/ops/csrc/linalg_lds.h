@@ -98,6 +98,58 @@ __device__ inline float dot4(const float* p, int sp, const float* q, int sq,
 }
 
 
+// ---------------------------------------------------------------------------
+// Pipelined variant: wave-specialized q loop with LDS flag counters
+// ---------------------------------------------------------------------------
+// Round-2 restructure of the C phase (TODO item 1, option b/c): the serial
+// 8x8 factor chain no longer waits for the 8x8 trtri or the full panel —
+// per sub-block q:
+//   wave 0:    spin(head[q-1]) -> 8x8 shfl factor(q) -> bump fact[q]
+//   wave 1:    spin(fact[q]) -> trtri8(q) from LDS (feeds ONLY the block
+//              inverse assembly, off the critical path) -> Vq
+//   waves 2-7: spin(fact[q]) -> panel rows [qb+8, bs) by forward
+//              substitution directly against L(q) (no vq needed!) ->
+//              group join -> trailing on the 16-row head slice -> bump
+//              head[q] (releases wave 0's next factor) -> trailing rest
+//              -> group join
+// Waves 1-7 also stream the previous panel's cross-block updates
+// (consumed only after the end-of-block s_barrier).  All spins are
+// bounded; on a bound hit or factor breakdown every wave falls through
+// to the barrier with *bad set.
+//
+// LDS flag block: the caller's 16-byte bad slot doubles as 4 ints —
+// [0] bad, [1] factor count, [2] head count, [3] group-B join count;
+// monotonic counters, workgroup-scope acquire/release.
+
+#define PIPE_SPIN_CAP 100000000
+
+__device__ inline int pipe_ld(int* p) {
+  return __hip_atomic_load(p, __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_WORKGROUP);
+}
+
+__device__ inline void pipe_st(int* p, int v) {
+  __hip_atomic_store(p, v, __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_WORKGROUP);
+}
+
+__device__ inline void pipe_add(int* p, int v) {
+  __hip_atomic_fetch_add(p, v, __ATOMIC_RELEASE,
+                         __HIP_MEMORY_SCOPE_WORKGROUP);
+}
+
+// spin until *p >= tgt (or bad); returns false on abort
+__device__ inline bool pipe_wait(int* p, int tgt, int* bad) {
+  int n = 0;
+  while (pipe_ld(p) < tgt) {
+    if (pipe_ld(bad)) return false;
+    if (++n > PIPE_SPIN_CAP) {
+      if (pipe_ld(bad) == 0) pipe_st(bad, 3);
+      return false;
+    }
+    __builtin_amdgcn_s_sleep(1);
+  }
+  return true;
+}
+
 // In place on the lower triangle of Abuf (k x k, row stride SA):
 // K -> L -> V = L^{-1} (lower).  Strict upper of Abuf is never touched.
 // log|K| accumulates into misc[0]; *bad set to 1 (indefinite) or 2
@@ -416,4 +468,323 @@ __device__ inline void chol_invert_lower(float* Abuf, float* Tbuf,
     __syncthreads();
   }
   if (jclk && tid == 0) jclk[5] += wall_clock64() - dt0;
+}
+
+// Pipelined C/D: same contract as chol_invert_lower (K -> L -> V in place,
+// logdet into misc[0], bad flag), restructured per the header note above.
+// flags = 4-int LDS block: [0] bad (shared with caller), [1] factor count,
+// [2] head count, [3] group-B join count — [1..3] reset per block here.
+// Used by expert_nll; laplace keeps the barrier version (smaller k budget,
+// different wave economics).
+__device__ inline void chol_invert_lower_pipe(
+    float* Abuf, float* Tbuf, const int k, const int SA, const int tid,
+    const int lane, int* flags, double* misc,
+    unsigned long long* jclk = nullptr) {
+  const int nblk = (k + NB - 1) / NB;
+  int* bad = flags;
+  int* FC = flags + 1;
+  int* FH = flags + 2;
+  int* FJ = flags + 3;
+  const int wave = tid >> 6;
+
+  for (int J = 0; J < nblk; ++J) {
+    const int jb = J * NB;
+    const int bs = min(NB, k - jb);
+    const int nq = (bs + 7) / 8;
+    float* D = Abuf + (size_t)jb * SA + jb;
+    const int pj = jb - NB;
+    if (tid == 0) { *FC = 0; *FH = 0; *FJ = 0; }
+    if (J > 0) {
+      // phase1: previous panel's rank-NB update to the FULL 32x32 block
+      // (lower incl. diag — every row this block's q loop touches)
+      for (int f = tid; f < bs * bs; f += WG) {
+        const int r = f / bs, c = f - r * bs;
+        if (c > r) continue;
+        const int i = jb + r, cc = jb + c;
+        Abuf[(size_t)i * SA + cc] -=
+            dotv(Abuf + (size_t)i * SA + pj,
+                 Abuf + (size_t)cc * SA + pj, 0, NB);
+      }
+    }
+    __syncthreads();
+    unsigned long long w0t0 = 0;
+    if (jclk && tid == 0) w0t0 = wall_clock64();
+
+    // ---------------- pipelined q loop (no barriers) -----------------
+    if (wave == 0) {
+      // wave 0: the factor chain only
+      for (int q = 0; q < nq; ++q) {
+        if (q > 0 && !pipe_wait(FH, q, bad)) break;
+        const int qb = q * 8;
+        const int sbs = min(8, bs - qb);
+        const int j = lane & 7;
+        float row[8];
+#pragma unroll
+        for (int c = 0; c < 8; ++c)
+          row[c] = (j < sbs && c < sbs && c <= j)
+                       ? D[(size_t)(qb + j) * SA + qb + c]
+                       : (c == j ? 1.f : 0.f);
+        bool ok = true, nonfin = false;
+        float mprod = 1.f;
+        int expsum = 0;
+        float myrs = 1.f;
+#pragma unroll
+        for (int ss = 0; ss < 8; ++ss) {
+          float l[8];
+#pragma unroll
+          for (int c = 0; c < 8; ++c) l[c] = __shfl(row[ss], c, 64);
+          const float piv = l[ss];
+          if (ss < sbs) {
+            const bool fin = isfinite(piv);
+            nonfin = nonfin || (ok && !fin);
+            ok = ok && fin && (piv > 0.f);
+            mprod *= __builtin_amdgcn_frexp_mantf(piv);
+            expsum += __builtin_amdgcn_frexp_expf(piv);
+          }
+          const float rs = rsqrtf(piv);
+          if (j == ss) myrs = rs;
+          const float w = (j > ss) ? rs * row[ss] : 0.f;
+#pragma unroll
+          for (int c = ss + 1; c < 8; ++c) row[c] -= (rs * l[c]) * w;
+        }
+#pragma unroll
+        for (int c = 0; c < 8; ++c) row[c] *= __shfl(myrs, c, 64);
+        if (lane < 8 && j < sbs) {
+#pragma unroll
+          for (int c = 0; c < 8; ++c)
+            if (c <= j) D[(size_t)(qb + j) * SA + qb + c] = row[c];
+        }
+        if (lane == 0) {
+          if (!ok) {
+            if (pipe_ld(bad) == 0) pipe_st(bad, nonfin ? 2 : 1);
+          } else {
+            misc[0] += (double)__logf(mprod)
+                       + (double)expsum * 0.6931471805599453;
+          }
+          pipe_add(FC, 1);
+        }
+        if (!ok) break;
+      }
+      if (jclk && tid == 0) jclk[2] += wall_clock64() - w0t0;
+    } else if (wave == 1) {
+      // wave 1: trtri8 into Vq (assembly input; off the critical path)
+      float* Vq = Tbuf;
+      for (int q = 0; q < nq; ++q) {
+        if (!pipe_wait(FC, q + 1, bad)) break;
+        const int qb = q * 8;
+        const int sbs = min(8, bs - qb);
+        if (lane < 8) {
+          const int j = lane;
+          float v[8];
+#pragma unroll
+          for (int i = 0; i < 8; ++i) v[i] = (i == j) ? 1.f : 0.f;
+          if (j < sbs) {
+            v[j] = __builtin_amdgcn_rcpf(
+                D[(size_t)(qb + j) * SA + qb + j]);
+            for (int i = j + 1; i < sbs; ++i) {
+              float sacc = 0.f;
+              for (int c = j; c < i; ++c)
+                sacc += D[(size_t)(qb + i) * SA + qb + c] * v[c];
+              v[i] = -sacc * __builtin_amdgcn_rcpf(
+                  D[(size_t)(qb + i) * SA + qb + i]);
+            }
+          }
+#pragma unroll
+          for (int i = 0; i < 8; ++i) Vq[q * 64 + i * 8 + j] = v[i];
+        }
+        // share of the previous panel's cross-block update (see below)
+        if (J > 0) {
+          const int t0r = jb + bs;
+          const int nrp = k - t0r;
+          const int ntri_rest = nrp * (nrp + 1) / 2;
+          for (int f = (tid - 64) + q * (WG - 64); f < nrp * bs;
+               f += (WG - 64) * nq) {
+            const int r = f / bs, c = f - r * bs;
+            const int i = t0r + r, cc = jb + c;
+            Abuf[(size_t)i * SA + cc] -=
+                dotv(Abuf + (size_t)i * SA + pj,
+                     Abuf + (size_t)cc * SA + pj, 0, NB);
+          }
+          for (int f = (tid - 64) + q * (WG - 64); f < ntri_rest;
+               f += (WG - 64) * nq) {
+            int a, b;
+            tri_decode(f, a, b);
+            const int i = t0r + a, c = t0r + b;
+            Abuf[(size_t)i * SA + c] -=
+                dotv(Abuf + (size_t)i * SA + pj,
+                     Abuf + (size_t)c * SA + pj, 0, NB);
+          }
+        }
+      }
+    } else {
+      // waves 2-7 (384 threads): panel substitution + trailing
+      const int gt = tid - 128;
+      for (int q = 0; q < nq; ++q) {
+        if (!pipe_wait(FC, q + 1, bad)) break;
+        const int qb = q * 8;
+        const int sbs = min(8, bs - qb);
+        const int p0 = qb + sbs;
+        const int pr = bs - p0;
+        // panel rows [p0, bs): solve p L(q)^T = a by forward
+        // substitution directly against L (no trtri dependency)
+        for (int r = p0 + gt; r < bs; r += 384) {
+          float p[8];
+          for (int c = 0; c < sbs; ++c)
+            p[c] = D[(size_t)r * SA + qb + c];
+          for (int c = 0; c < sbs; ++c) {
+            float s = p[c];
+            for (int t = 0; t < c; ++t)
+              s -= p[t] * D[(size_t)(qb + c) * SA + qb + t];
+            p[c] = s * __builtin_amdgcn_rcpf(
+                D[(size_t)(qb + c) * SA + qb + c]);
+          }
+          for (int c = 0; c < sbs; ++c)
+            D[(size_t)r * SA + qb + c] = p[c];
+        }
+        if (lane == 0) pipe_add(FJ, 1);                 // join A
+        if (!pipe_wait(FJ, 18 * q + 6, bad)) break;
+        // trailing, head slice first (rows [p0, p0+16)): releases the
+        // next factor as soon as its rows are complete
+        const int hend = min(p0 + 16, bs);
+        const int hr = hend - p0;
+        const int nth = hr * (hr + 1) / 2;
+        for (int f = gt; f < nth; f += 384) {
+          int a, b;
+          tri_decode(f, a, b);
+          const int r = p0 + a, c = p0 + b;
+          float s = 0.f;
+          for (int t = 0; t < sbs; ++t)
+            s += D[(size_t)r * SA + qb + t] * D[(size_t)c * SA + qb + t];
+          D[(size_t)r * SA + c] -= s;
+        }
+        if (lane == 0) pipe_add(FJ, 1);                 // join B
+        if (!pipe_wait(FJ, 18 * q + 12, bad)) break;
+        if (gt == 0) pipe_add(FH, 1);
+        // trailing rest: the remaining pairs of the [p0, bs) triangle
+        // (flat indices continue past the head block: tri_decode
+        // enumerates rows ascending, so head pairs are exactly the
+        // first nth indices)
+        const int ntri_all = pr * (pr + 1) / 2;
+        for (int f = nth + gt; f < ntri_all; f += 384) {
+          int a, b;
+          tri_decode(f, a, b);
+          const int r = p0 + a, c = p0 + b;
+          float s = 0.f;
+          for (int t = 0; t < sbs; ++t)
+            s += D[(size_t)r * SA + qb + t] * D[(size_t)c * SA + qb + t];
+          D[(size_t)r * SA + c] -= s;
+        }
+        // share of the previous panel's cross-block update
+        if (J > 0) {
+          const int t0r = jb + bs;
+          const int nrp = k - t0r;
+          const int ntri_rest = nrp * (nrp + 1) / 2;
+          for (int f = (tid - 64) + q * (WG - 64); f < nrp * bs;
+               f += (WG - 64) * nq) {
+            const int r = f / bs, c = f - r * bs;
+            const int i = t0r + r, cc = jb + c;
+            Abuf[(size_t)i * SA + cc] -=
+                dotv(Abuf + (size_t)i * SA + pj,
+                     Abuf + (size_t)cc * SA + pj, 0, NB);
+          }
+          for (int f = (tid - 64) + q * (WG - 64); f < ntri_rest;
+               f += (WG - 64) * nq) {
+            int a, b;
+            tri_decode(f, a, b);
+            const int i = t0r + a, c = t0r + b;
+            Abuf[(size_t)i * SA + c] -=
+                dotv(Abuf + (size_t)i * SA + pj,
+                     Abuf + (size_t)c * SA + pj, 0, NB);
+          }
+        }
+        if (lane == 0) pipe_add(FJ, 1);                 // join C
+        if (!pipe_wait(FJ, 18 * q + 18, bad)) break;
+      }
+    }
+    __syncthreads();
+    if (*bad) break;
+
+    // ---- assemble V_JJ (32x32 inverse) from the 8x8 inverses ---------
+    {
+      const int nq2 = (bs + 7) / 8;
+      float* Vq = Tbuf;
+      float* TS = Tbuf + 256;
+      for (int Jq = nq2 - 1; Jq >= 0; --Jq) {
+        const int jb8 = Jq * 8;
+        const int nblks = nq2 - 1 - Jq;
+        for (int f = tid; f < nblks * 64; f += WG) {
+          const int blk = f >> 6;
+          const int ib = (Jq + 1 + blk) * 8;
+          const int i = (f >> 3) & 7, j = f & 7;
+          const int gi = ib + i;
+          float u = 0.f;
+          if (gi < bs && jb8 + j < bs) {
+            u = dotm(D + (size_t)gi * SA, D + jb8 + j, SA,
+                     jb8 + 8, min(gi + 1, bs));
+          }
+          TS[f] = u;
+        }
+        __syncthreads();
+        for (int f = tid; f < nblks * 64; f += WG) {
+          const int blk = f >> 6;
+          const int ib = (Jq + 1 + blk) * 8;
+          const int i = (f >> 3) & 7, j = f & 7;
+          const int gi = ib + i;
+          if (gi >= bs || jb8 + j >= bs) continue;
+          float sacc = 0.f;
+          const float* ts = TS + blk * 64 + i * 8;
+          const float* vv = Vq + Jq * 64;
+#pragma unroll
+          for (int t = 0; t < 8; ++t) sacc += ts[t] * vv[t * 8 + j];
+          D[(size_t)gi * SA + jb8 + j] = -sacc;
+        }
+        for (int f = tid; f < 64; f += WG) {
+          const int i = (f >> 3) & 7, j = f & 7;
+          const int gi = jb8 + i, gj = jb8 + j;
+          if (gi < bs && gj <= gi)
+            D[(size_t)gi * SA + gj] = Vq[Jq * 64 + f];
+        }
+        __syncthreads();
+      }
+    }
+
+    const int t0 = jb + bs;
+    const int nr = k - t0;
+    if (nr > 0) {
+      // C2: cross-block panel solve against the inverted diagonal
+      for (int f = tid; f < nr * bs; f += WG) {
+        int r = f / bs, c = f - r * bs;
+        Tbuf[r * 36 + c] = Abuf[(size_t)(t0 + r) * SA + jb + c];
+      }
+      __syncthreads();
+      for (int f = tid; f < nr * bs; f += WG) {
+        int r = f / bs, c = f - r * bs;
+        Abuf[(size_t)(t0 + r) * SA + jb + c] =
+            dotv(Tbuf + r * 36, D + c * SA, 0, c + 1);
+      }
+      __syncthreads();
+    }
+  }
+
+  if (*bad) return;
+  // ---- D: off-diagonal triangular inverse (same as the barrier path)
+  for (int J = nblk - 2; J >= 0; --J) {
+    const int jb = J * NB;
+    const int bs = NB;
+    const int t0 = jb + bs;
+    const int nr = k - t0;
+    for (int f = tid; f < nr * bs; f += WG) {
+      int r = f / bs, t = f - r * bs;
+      const int row = t0 + r;
+      Tbuf[r * 36 + t] = dotm(Abuf + (size_t)row * SA,
+                              Abuf + jb + t, SA, t0, row + 1);
+    }
+    __syncthreads();
+    for (int f = tid; f < nr * bs; f += WG) {
+      int r = f / bs, j = f - r * bs;
+      Abuf[(size_t)(t0 + r) * SA + jb + j] =
+          -dotm(Tbuf + r * 36, Abuf + (size_t)jb * SA + jb + j, SA, j, bs);
+    }
+    __syncthreads();
+  }
 }
