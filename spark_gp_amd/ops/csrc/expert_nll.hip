@@ -194,8 +194,8 @@ PH(2);
   // ---- C/D: in-place blocked Cholesky + triangular inverse ---------
   // (shared machinery: linalg_lds.h)  A: lower K -> V = L^-1; upper Kb
   // cache untouched; log|K| into misc[0]; bad flag on fp32 breakdown.
-  chol_invert_lower_pipe(S.A, S.T, k, SA, tid, lane, S.bad, S.misc,
-                         out_clk ? out_clk + (size_t)e * 20 + 12 : nullptr);
+  chol_invert_lower(S.A, S.T, k, SA, tid, lane, S.bad, S.misc,
+                    out_clk ? out_clk + (size_t)e * 20 + 12 : nullptr);
   if (*S.bad) {
     if (tid == 0) {
       out_bad[e] = *S.bad;     // 1: indefinite, 2: non-finite iterate

@@ -101,9 +101,16 @@ __device__ inline float dot4(const float* p, int sp, const float* q, int sq,
 // ---------------------------------------------------------------------------
 // Pipelined variant: wave-specialized q loop with LDS flag counters
 // ---------------------------------------------------------------------------
-// Round-2 restructure of the C phase (TODO item 1, option b/c): the serial
-// 8x8 factor chain no longer waits for the 8x8 trtri or the full panel —
-// per sub-block q:
+// Round-2 restructure of the C phase (TODO item 1, option b/c) — kept as
+// a documented NEGATIVE RESULT: measured 133 us for the C phase at k=100
+// vs 88 us for the barrier version (E=20000, launch 7.32 vs 5.58 ms).
+// The group-B panel substitution is a serial 8-step LDS-latency chain
+// (~1.5-2 us/step regardless of thread count) that replaces the trtri on
+// the critical cycle, and the per-step join cadence adds flag latency;
+// the barrier version's all-thread panel GEMM against the inverted 8x8
+// is faster than anything that avoids the trtri.  Not referenced by any
+// kernel (inline => not emitted); kept for the design record.
+// Per sub-block q:
 //   wave 0:    spin(head[q-1]) -> 8x8 shfl factor(q) -> bump fact[q]
 //   wave 1:    spin(fact[q]) -> trtri8(q) from LDS (feeds ONLY the block
 //              inverse assembly, off the critical path) -> Vq
