@@ -95,11 +95,17 @@ def main():
 
     # fixed synthetic design matrix: y = sin(2 sum x) + noise, x ~ U[0,1)^d
     # (non-degenerate GP target; see data/synthetic.py for why the reference
-    # harness's sin(sum x/1000) is a constant-predictor corner case)
-    X, y = shard_performance_benchmark_data(args.rows, args.dim, rank, world,
-                                            seed=args.seed)
-    Xt = torch.as_tensor(X, device=device)
-    yt = torch.as_tensor(y, device=device)
+    # harness's sin(sum x/1000) is a constant-predictor corner case).
+    # On GPU the shard is generated in place by the K19 Philox kernel.
+    if device.type == "cuda":
+        from spark_gp_amd.data import shard_benchmark_data_device
+        Xt, yt = shard_benchmark_data_device(args.rows, args.dim, rank,
+                                             world, seed=args.seed)
+    else:
+        X, y = shard_performance_benchmark_data(args.rows, args.dim, rank,
+                                                world, seed=args.seed)
+        Xt = torch.as_tensor(X, device=device)
+        yt = torch.as_tensor(y, device=device)
     dim = args.dim
 
     def make_gp():

@@ -24,6 +24,7 @@ ext = CUDAExtension(
         "spark_gp_amd/ops/csrc/cross_syrk.hip",
         "spark_gp_amd/ops/csrc/laplace.hip",
         "spark_gp_amd/ops/csrc/big_chol.hip",
+        "spark_gp_amd/ops/csrc/synth.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3"],

@@ -83,6 +83,22 @@ def mnist_like_binary(n_rows: int = 11769, side: int = 28, modes: int = 8,
     return X.astype(dtype), ys
 
 
+def shard_benchmark_data_device(n_total: int, d: int, rank: int,
+                                world_size: int, seed: int = 13,
+                                noise_sd: float = 0.1):
+    """This rank's shard of the flagship benchmark data, generated ON the
+    GPU by the K19 Philox4x32-10 kernel (synth.hip) — same distribution as
+    ``benchmark_regression_data`` (X ~ U[0,1)^d, y = sin(2 sum x) + noise)
+    without the host-side generation + H2D that costs minutes and >100 GB
+    of host RAM at the 100M x 128 config."""
+    base = n_total // world_size
+    rem = n_total % world_size
+    n_local = base + (1 if rank < rem else 0)
+    from spark_gp_amd import _hip_ext
+    return _hip_ext.synth_regression(n_local, d, int(seed + 1009 * rank),
+                                     float(noise_sd))
+
+
 def shard_performance_benchmark_data(n_total: int, d: int, rank: int,
                                      world_size: int, seed: int = 13,
                                      dtype=np.float32):

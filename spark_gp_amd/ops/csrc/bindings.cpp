@@ -32,6 +32,11 @@ extern "C" hipError_t launch_fused_laplace_evidence(
     double* out_logz, double* out_grad, hipStream_t stream,
     size_t* lds_used);
 
+extern "C" hipError_t launch_synth_regression(float* X, float* y, long n,
+                                              int d, unsigned long long seed,
+                                              float noise_sd,
+                                              hipStream_t stream);
+
 extern "C" hipError_t launch_dpotrf_diag(double* A, long m, int jb,
                                          double* Vout, int* bad,
                                          hipStream_t stream);
@@ -394,7 +399,23 @@ torch::Tensor dgemm64(torch::Tensor A, torch::Tensor B, bool ta, bool tb) {
   return C;
 }
 
+// K19: device-side synthetic benchmark data (Philox4x32-10)
+std::vector<torch::Tensor> synth_regression(int64_t n, int64_t d,
+                                            int64_t seed, double noise_sd) {
+  auto dev = torch::TensorOptions().dtype(torch::kFloat32)
+                 .device(torch::kCUDA);
+  auto X = torch::empty({n, d}, dev);
+  auto y = torch::empty({n}, dev);
+  check_hip(launch_synth_regression(X.data_ptr<float>(), y.data_ptr<float>(),
+                                    n, (int)d, (unsigned long long)seed,
+                                    (float)noise_sd, current_stream()),
+            "synth_regression");
+  return {X, y};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("synth_regression", &synth_regression,
+          "device-side Philox benchmark data (K19)");
   mod.def("dpotrf64", &dpotrf64,
           "blocked fp64 Cholesky (K13), in place, 64-padded");
   mod.def("dchol_solve64", &dchol_solve64,

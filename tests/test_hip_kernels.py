@@ -398,3 +398,26 @@ def test_fused_laplace_evidence_vs_torch(dev, ext, base):
     assert nll2 == pytest.approx(nll_hip, rel=1e-6)
     np.testing.assert_allclose(grad2, grad_hip, rtol=1e-6,
                                atol=1e-6 * np.abs(grad_hip).max())
+
+
+def test_synth_regression_device_gen(dev, ext):
+    """K19: the Philox device generator's distribution must match the host
+    benchmark data: X ~ U[0,1)^d, y = sin(2 sum x) + 0.1 N(0,1); counter-
+    based => bitwise deterministic per (seed, row)."""
+    X, y = ext.synth_regression(200_000, 8, 13, 0.1)
+    assert X.shape == (200_000, 8) and y.shape == (200_000,)
+    mx = X.mean().item()
+    vx = X.var().item()
+    assert abs(mx - 0.5) < 2e-3, mx
+    assert abs(vx - 1.0 / 12.0) < 1e-3, vx
+    resid = y - torch.sin(2.0 * X.sum(-1))
+    assert abs(resid.mean().item()) < 2e-3
+    assert abs(resid.std().item() - 0.1) < 2e-3
+    # determinism + seed sensitivity
+    X2, y2 = ext.synth_regression(200_000, 8, 13, 0.1)
+    assert torch.equal(X, X2) and torch.equal(y, y2)
+    X3, _ = ext.synth_regression(200_000, 8, 14, 0.1)
+    assert not torch.equal(X, X3)
+    # no pathological correlation between adjacent columns
+    c = torch.corrcoef(X[:, :2].T)[0, 1].abs().item()
+    assert c < 0.01, c
