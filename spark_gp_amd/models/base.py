@@ -252,6 +252,9 @@ class GaussianProcessCommons(GaussianProcessParams):
 
         evals = [0, 0.0]
 
+        import os as _os
+        trace = _os.environ.get("SPARK_GP_AMD_TRACE_OBJECTIVE") == "1"
+
         def objective(theta: np.ndarray) -> Tuple[float, np.ndarray]:
             te = time.perf_counter()
             nll, grad = local_obj(theta)
@@ -259,6 +262,9 @@ class GaussianProcessCommons(GaussianProcessParams):
             buf = comm.allreduce_np(buf)
             evals[0] += 1
             evals[1] += time.perf_counter() - te
+            if trace and comm.rank == 0:
+                print(f"[obj] eval={evals[0]} nll={float(buf[0]):.10g}",
+                      flush=True)
             return float(buf[0]), buf[1:]
 
         t0 = time.perf_counter()
