@@ -56,3 +56,20 @@ torch.cuda.synchronize()
 dt = (time.perf_counter() - t0) / reps
 print(f"{dt * 1e3:.3f} ms/call  ->  {flops / dt / 1e12:.1f} TF bf16 "
       f"({flops / 3 / dt / 1e12:.1f} TF fp32-equivalent)")
+
+# optional split_k sweep: shorter per-block k-ranges shrink the live
+# column-window working set toward L3/L2 residency (the round-1 TCC
+# analysis: 38% hit, ~6x compulsory traffic from drifting tiles) at the
+# price of split_k-fold output atomic traffic — measure the trade.
+if "--sweep" in sys.argv:
+    for sk in (split_k, 8, 16, 32, 64, 128, 256):
+        for _ in range(2):
+            ext.syrk_bf16_acc(KcT, KlT, KK, sk)
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(10):
+            ext.syrk_bf16_acc(KcT, KlT, KK, sk)
+        torch.cuda.synchronize()
+        dt = (time.perf_counter() - t0) / 10
+        print(f"  split_k={sk:4d}: {dt * 1e3:8.3f} ms  "
+              f"{flops / dt / 1e12:6.1f} TF bf16")
