@@ -228,9 +228,12 @@ def kmn_knm_and_kmny(kernel: Kernel, active: torch.Tensor,
     KK = torch.zeros(m, m, dtype=torch.float32, device=X.device)
     ntile = (m + 255) // 256
     tiles = ntile * (ntile + 1) // 2
-    # fill the machine: the 512-thread/80-KB-LDS SYRK runs 1 block/CU ->
-    # 256 slots; more slices than that only adds atomic traffic
-    split_k = max(1, min(64, 256 // tiles))
+    # split_k: measured sweep (scripts/bench_syrk.py --sweep, r2): short
+    # per-block k-ranges shrink the drifting column-window working set
+    # toward L3 residency and beat the old fill-the-machine rule — best
+    # 16 at m=1000 (433 vs 367 TF), 8 at m=8192 (516 vs 495 TF); beyond
+    # that the split_k-fold output atomic traffic wins out.
+    split_k = max(1, min(16, round(4096 / tiles)))
     for s in range(0, n, chunk_rows):
         e = min(n, s + chunk_rows)
         # the cross kernel also emits transposed (k-contiguous) copies so
