@@ -281,27 +281,25 @@ __device__ inline void chol_invert_lower(float* Abuf, float* Tbuf,
             for (int c = 0; c < 8; ++c)
               if (c <= j) D[(size_t)(qb + j) * SA + qb + c] = row[c];
           }
-          // trtri8: V column j by forward substitution, L read BACK from
-          // LDS (static addresses issue eagerly and pipeline; the old
-          // per-i __shfl row gathers were 64 ds_bpermutes whose ~50-cycle
-          // latencies sat inside the dependent substitution chain —
-          // replacing them cut the factor+trtri wave-0 span measurably).
-          // Lanes >= 8 skip entirely (they hold duplicate rows).
+          // trtri8: V column j by forward substitution; row i of L is
+          // gathered as rowi[c] = lane i's row[c]
           float v[8];
 #pragma unroll
           for (int i = 0; i < 8; ++i) v[i] = (i == j) ? myrs : 0.f;
-          if (lane < 8) {
 #pragma unroll
-            for (int i = 1; i < 8; ++i) {
-              if (j < i && i < sbs) {
-                const float* Li = D + (size_t)(qb + i) * SA + qb;
-                float sacc = 0.f;
+          for (int i = 0; i < 8; ++i) {
+            float rowi[8];
 #pragma unroll
-                for (int c = 0; c < 8; ++c)
-                  if (c >= j && c < i) sacc += Li[c] * v[c];
-                v[i] = -sacc * __builtin_amdgcn_rcpf(Li[i]);
-              }
+            for (int c = 0; c < 8; ++c) rowi[c] = __shfl(row[c], i, 64);
+            if (j < i) {
+              float sacc = 0.f;
+#pragma unroll
+              for (int c = 0; c < 8; ++c)
+                if (c >= j && c < i) sacc += rowi[c] * v[c];
+              v[i] = -sacc * __builtin_amdgcn_rcpf(rowi[i]);
             }
+          }
+          if (lane < 8) {
 #pragma unroll
             for (int i = 0; i < 8; ++i) Vq[q * 64 + i * 8 + j] = v[i];
           }
