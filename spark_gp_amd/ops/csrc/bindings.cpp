@@ -21,6 +21,13 @@ extern "C" hipError_t launch_syrk_bf16(const void* KcT, const void* KlT,
                                        int c, int m, int cpitch, int split_k,
                                        float* KK, hipStream_t stream);
 
+extern "C" hipError_t launch_syrk_bf16_sync(const void* KcT, const void* KlT,
+                                            int c, int m, int cpitch,
+                                            const int* tiles, int nb,
+                                            int kpb, int* phase_ctr,
+                                            int nactive, float* KK,
+                                            hipStream_t stream);
+
 extern "C" hipError_t launch_colsum_gemv(const void* Kc, const float* y,
                                          int c, int m, double* Ky,
                                          hipStream_t stream);
@@ -190,6 +197,37 @@ void syrk_bf16_acc(torch::Tensor KcT, c10::optional<torch::Tensor> KlT,
                              c, (int)split_k, KK.data_ptr<float>(),
                              current_stream()),
             "syrk_bf16");
+}
+
+// k-synchronized SYRK (large m): tiles [nb, 2] int32 (ti, tj; -1 pads),
+// one block per tile, accumulators held across all k; see cross_syrk.hip.
+void syrk_bf16_sync_acc(torch::Tensor KcT, c10::optional<torch::Tensor> KlT,
+                        torch::Tensor KK, torch::Tensor tiles,
+                        int64_t kpb, int64_t nactive) {
+  TORCH_CHECK(KcT.is_cuda() && KcT.dtype() == torch::kBFloat16 &&
+              KcT.dim() == 2 && KcT.is_contiguous());
+  TORCH_CHECK(KK.is_cuda() && KK.dtype() == torch::kFloat32 && KK.dim() == 2);
+  TORCH_CHECK(tiles.is_cuda() && tiles.dtype() == torch::kInt32 &&
+              tiles.dim() == 2 && tiles.size(1) == 2 && tiles.is_contiguous());
+  const int m = KcT.size(0), c = KcT.size(1);
+  TORCH_CHECK(KK.size(0) == m && KK.size(1) == m);
+  torch::Tensor Klc;
+  if (KlT.has_value()) {
+    TORCH_CHECK(KlT->sizes() == KcT.sizes() &&
+                KlT->dtype() == torch::kBFloat16);
+    Klc = KlT->contiguous();
+  }
+  const int kblocks = (c + 31) / 32;
+  const int nphases = (kblocks + (int)kpb - 1) / (int)kpb;
+  auto ctr = torch::zeros({nphases},
+                          torch::TensorOptions().dtype(torch::kInt32)
+                              .device(KK.device()));
+  check_hip(launch_syrk_bf16_sync(
+                KcT.data_ptr(), KlT.has_value() ? Klc.data_ptr() : nullptr,
+                c, m, c, tiles.data_ptr<int>(), (int)tiles.size(0),
+                (int)kpb, ctr.data_ptr<int>(), (int)nactive,
+                KK.data_ptr<float>(), current_stream()),
+            "syrk_bf16_sync");
 }
 
 void colsum_gemv_acc(torch::Tensor Kc, torch::Tensor y, torch::Tensor Ky) {
@@ -438,6 +476,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           pybind11::arg("X"), pybind11::arg("A"), pybind11::arg("s2v"),
           pybind11::arg("amp"), pybind11::arg("bf16_out"),
           pybind11::arg("hilo"), pybind11::arg("want_t") = false);
+  mod.def("syrk_bf16_sync_acc", &syrk_bf16_sync_acc,
+          "k-synchronized persistent SYRK for large m (CDNA4)");
   mod.def("syrk_bf16_acc", &syrk_bf16_acc,
           "KK += K K^T of the transposed block KcT [m, c]; bf16 MFMA, "
           "fp32 accumulate (CDNA4)");

@@ -328,6 +328,173 @@ syrk_bf16_kernel(const bf16* __restrict__ KcT,  // [m, cpitch] hi^T
 // leaving KK fully populated.
 
 // ---------------------------------------------------------------------------
+// syrk_bf16_sync: k-SYNCHRONIZED variant for large m (round-2, TODO item 1)
+// ---------------------------------------------------------------------------
+// At m >= 4096 the plain kernel's co-scheduled tiles drift apart in k, so
+// no XCD's L2 ever holds the column windows its tiles are reading (round-1
+// TCC analysis: 38% hit, ~6x compulsory HBM traffic).  This variant keeps
+// every live tile on the SAME k-phase: one block owns ONE output tile for
+// the whole launch (accumulators never leave AGPRs), the host passes a
+// tile list clustered so each XCD's blocks share row/column windows, and
+// blocks pace each other between k-phases through a per-phase arrival
+// counter.  The pacing spin is BOUNDED and best-effort: there is no data
+// dependency between blocks, so a timeout only loses locality, never
+// correctness — no grid-residency assumption, no deadlock risk.
+
+template <bool HILO>
+__global__ void __launch_bounds__(SY_WG, 2)
+syrk_bf16_sync_kernel(const bf16* __restrict__ KcT,   // [m, cpitch] hi^T
+                      const bf16* __restrict__ KlT,   // [m, cpitch] lo^T
+                      const int c, const int m, const int cpitch,
+                      const int* __restrict__ tiles,  // [nb, 2]; -1 = idle
+                      const int kpb,                  // k-blocks per phase
+                      int* __restrict__ phase_ctr,    // [nphases], zeroed
+                      const int nactive,
+                      float* __restrict__ KK) {
+  const int ti = tiles[2 * blockIdx.x];
+  const int tj = tiles[2 * blockIdx.x + 1];
+  if (ti < 0) return;
+  const int i0 = ti * SY_CT, j0 = tj * SY_CT;
+  const int kblocks = (c + SY_BK - 1) / SY_BK;
+
+  __shared__ __align__(16) bf16 lt[SY_CT * SY_STR];
+  __shared__ __align__(16) bf16 rt[SY_CT * SY_STR];
+  __shared__ __align__(16) bf16 ltl[HILO ? SY_CT * SY_STR : 1];
+  __shared__ __align__(16) bf16 rtl[HILO ? SY_CT * SY_STR : 1];
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int wr = (wave >> 1) * 64;
+  const int wc = (wave & 1) * 128;
+
+  f32x4 acc[4][8];
+#pragma unroll
+  for (int a = 0; a < 4; ++a)
+#pragma unroll
+    for (int b = 0; b < 8; ++b) acc[a][b] = {0.f, 0.f, 0.f, 0.f};
+
+  const int l16 = lane & 15;
+  const int kgrp = lane >> 4;
+  const bool vec_ok = (cpitch % 8 == 0);
+
+  auto load4 = [&](const bf16* mat, int c0, int col, int gk0) -> uint4 {
+    uint4 v;
+    bf16* vals = (bf16*)&v;
+    const int gc = c0 + col;
+    if (gc < m && vec_ok && gk0 + 8 <= c) {
+      v = *(const uint4*)(mat + (size_t)gc * cpitch + gk0);
+    } else if (gc < m) {
+#pragma unroll
+      for (int u = 0; u < 8; ++u)
+        vals[u] = (gk0 + u < c) ? mat[(size_t)gc * cpitch + gk0 + u]
+                                : (bf16)0.f;
+    } else {
+#pragma unroll
+      for (int u = 0; u < 8; ++u) vals[u] = (bf16)0.f;
+    }
+    return v;
+  };
+
+  uint4 sv[2][4];
+  auto load_all = [&](int kb) {
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      const int f = tid + it * SY_WG;
+      const int col = f >> 2;
+      const int gk0 = kb * SY_BK + (f & 3) * 8;
+      sv[it][0] = load4(KcT, i0, col, gk0);
+      sv[it][1] = load4(KcT, j0, col, gk0);
+      if (HILO) {
+        sv[it][2] = load4(KlT, i0, col, gk0);
+        sv[it][3] = load4(KlT, j0, col, gk0);
+      }
+    }
+  };
+  auto write_all = [&]() {
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      const int f = tid + it * SY_WG;
+      const int o = (f >> 2) * SY_STR + (f & 3) * 8;
+      *(uint4*)&lt[o] = sv[it][0];
+      *(uint4*)&rt[o] = sv[it][1];
+      if (HILO) {
+        *(uint4*)&ltl[o] = sv[it][2];
+        *(uint4*)&rtl[o] = sv[it][3];
+      }
+    }
+  };
+  auto mfma_pass = [&](const bf16* at, const bf16* bt) {
+    bf16x8 fa[4], fb[4];
+    const int ko = kgrp * 8;
+#pragma unroll
+    for (int a = 0; a < 4; ++a)
+      fa[a] = *(const bf16x8*)&at[(wr + a * 16 + l16) * SY_STR + ko];
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+#pragma unroll
+      for (int b = 0; b < 4; ++b)
+        fb[b] = *(const bf16x8*)&bt[(wc + (h * 4 + b) * 16 + l16) * SY_STR
+                                    + ko];
+#pragma unroll
+      for (int a = 0; a < 4; ++a)
+#pragma unroll
+        for (int b = 0; b < 4; ++b)
+          acc[a][h * 4 + b] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              fa[a], fb[b], acc[a][h * 4 + b], 0, 0, 0);
+    }
+  };
+
+  load_all(0);
+  const int nphases = (kblocks + kpb - 1) / kpb;
+  for (int p = 0; p < nphases; ++p) {
+    const int kb1 = min(kblocks, (p + 1) * kpb);
+    for (int kb = p * kpb; kb < kb1; ++kb) {
+      write_all();
+      __syncthreads();
+      if (kb + 1 < kblocks) load_all(kb + 1);
+      mfma_pass(lt, rt);
+      if (HILO) {
+        mfma_pass(lt, rtl);
+        mfma_pass(ltl, rt);
+      }
+      __syncthreads();
+    }
+    // best-effort pacing: arrive, then wait (bounded) for the cohort.
+    // Relaxed atomics only — nothing is communicated, so no fences.
+    if (p + 1 < nphases) {
+      if (tid == 0) {
+        __hip_atomic_fetch_add(&phase_ctr[p], 1, __ATOMIC_RELAXED,
+                               __HIP_MEMORY_SCOPE_AGENT);
+        int spins = 0;
+        while (__hip_atomic_load(&phase_ctr[p], __ATOMIC_RELAXED,
+                                 __HIP_MEMORY_SCOPE_AGENT) < nactive &&
+               ++spins < 1500)
+          __builtin_amdgcn_s_sleep(8);
+      }
+      __syncthreads();
+    }
+  }
+
+  // epilogue: single owner per element per launch -> plain accumulate
+  const int crow_base = (lane >> 4) * 4;
+  const int ccol = lane & 15;
+#pragma unroll
+  for (int a = 0; a < 4; ++a)
+#pragma unroll
+    for (int b = 0; b < 8; ++b)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int gi = i0 + wr + a * 16 + crow_base + r;
+        const int gj = j0 + wc + b * 16 + ccol;
+        if (gi < m && gj < m) {
+          KK[(size_t)gi * m + gj] += acc[a][b][r];
+          if (ti != tj) KK[(size_t)gj * m + gi] += acc[a][b][r];
+        }
+      }
+}
+
+// ---------------------------------------------------------------------------
 // colsum_gemv: Ky[m] += Kc^T y (fp64 accumulate per block, one atomic per
 // column)
 // ---------------------------------------------------------------------------
@@ -365,6 +532,24 @@ extern "C" hipError_t launch_cross_kernel_tile(
                      out_is_bf16 ? (bf16*)out_lo : nullptr,
                      (bf16*)out_t, (bf16*)out_lo_t,
                      out_is_bf16 ? nullptr : (float*)out);
+  return hipGetLastError();
+}
+
+extern "C" hipError_t launch_syrk_bf16_sync(const void* KcT, const void* KlT,
+                                            int c, int m, int cpitch,
+                                            const int* tiles, int nb,
+                                            int kpb, int* phase_ctr,
+                                            int nactive, float* KK,
+                                            hipStream_t stream) {
+  if (KlT) {
+    hipLaunchKernelGGL((syrk_bf16_sync_kernel<true>), dim3(nb), dim3(SY_WG),
+                       0, stream, (const bf16*)KcT, (const bf16*)KlT, c, m,
+                       cpitch, tiles, kpb, phase_ctr, nactive, KK);
+  } else {
+    hipLaunchKernelGGL((syrk_bf16_sync_kernel<false>), dim3(nb), dim3(SY_WG),
+                       0, stream, (const bf16*)KcT, nullptr, c, m,
+                       cpitch, tiles, kpb, phase_ctr, nactive, KK);
+  }
   return hipGetLastError();
 }
 

@@ -195,6 +195,56 @@ def _s2_vector(cs: CompiledKernel, theta: np.ndarray, d: int, device):
     return (s * s).to(torch.float32)
 
 
+_SYNC_TILE_CACHE: dict = {}
+
+
+def _syrk_sync_tiles(m: int, device):
+    """XCD-clustered tile lists for the k-synchronized SYRK.
+
+    One block owns one 256x256 output tile for a whole launch, so a launch
+    carries at most 224 tiles (blocks beyond the resident set would break
+    the k-cohort).  The dispatcher places block b on XCD b%8 (observed,
+    speed-only), so each launch list interleaves 8 per-XCD sequences whose
+    tiles are sorted by 2x4-tile patches — an XCD's resident tiles then
+    share a handful of 256-row/column operand windows that fit its 4 MB L2
+    at the 256-k phase width."""
+    key = (m, str(device))
+    if key in _SYNC_TILE_CACHE:
+        return _SYNC_TILE_CACHE[key]
+    ntile = (m + 255) // 256
+    tiles = [(ti, tj) for ti in range(ntile) for tj in range(ti + 1)]
+    tiles.sort(key=lambda t: (t[0] // 2, t[1] // 4))
+    launches = []
+    CAP = 224
+    for s in range(0, len(tiles), CAP):
+        grp = tiles[s:s + CAP]
+        per = (len(grp) + 7) // 8
+        L = [(-1, -1)] * (per * 8)
+        for x in range(8):
+            seg = grp[x * per:(x + 1) * per]
+            for j, t in enumerate(seg):
+                L[j * 8 + x] = t
+        tt = torch.tensor(L, dtype=torch.int32, device=device)
+        launches.append((tt, len(grp)))
+    _SYNC_TILE_CACHE[key] = launches
+    return launches
+
+
+def _syrk_dispatch(KcT, KlT, KK, m: int):
+    import os
+    if m >= 4096 and os.environ.get("SPARK_GP_AMD_SYRK_SYNC", "1") == "1":
+        for tt, nact in _syrk_sync_tiles(m, KK.device):
+            ext.syrk_bf16_sync_acc(KcT, KlT, KK, tt, 8, nact)
+        return
+    ntile = (m + 255) // 256
+    tiles = ntile * (ntile + 1) // 2
+    # split_k: measured sweep (scripts/bench_syrk.py --sweep, r2) — see
+    # PROFILES.md; shorter per-block k-ranges keep the drifting column
+    # windows closer to L3 residency
+    split_k = max(1, min(16, round(4096 / tiles)))
+    ext.syrk_bf16_acc(KcT, KlT, KK, split_k)
+
+
 def kmn_knm_and_kmny(kernel: Kernel, active: torch.Tensor,
                      X: torch.Tensor, y: torch.Tensor,
                      chunk_rows: int = 131072, precision: str = "fp64"
@@ -226,14 +276,6 @@ def kmn_knm_and_kmny(kernel: Kernel, active: torch.Tensor,
     # MFMA SYRK keeps input-quantization error at fp32 class while running
     # at bf16 matrix-core rate
     KK = torch.zeros(m, m, dtype=torch.float32, device=X.device)
-    ntile = (m + 255) // 256
-    tiles = ntile * (ntile + 1) // 2
-    # split_k: measured sweep (scripts/bench_syrk.py --sweep, r2): short
-    # per-block k-ranges shrink the drifting column-window working set
-    # toward L3 residency and beat the old fill-the-machine rule — best
-    # 16 at m=1000 (433 vs 367 TF), 8 at m=8192 (516 vs 495 TF); beyond
-    # that the split_k-fold output atomic traffic wins out.
-    split_k = max(1, min(16, round(4096 / tiles)))
     for s in range(0, n, chunk_rows):
         e = min(n, s + chunk_rows)
         # the cross kernel also emits transposed (k-contiguous) copies so
@@ -241,7 +283,7 @@ def kmn_knm_and_kmny(kernel: Kernel, active: torch.Tensor,
         Kc, Kl, KcT, KlT = ext.cross_kernel_tile(X[s:e].contiguous(), act32,
                                                  s2, float(C), True, True,
                                                  True)
-        ext.syrk_bf16_acc(KcT, KlT, KK, split_k)
+        _syrk_dispatch(KcT, KlT, KK, m)
         yc = y32[s:e].contiguous()
         ext.colsum_gemv_acc(Kc, yc, Ky)
         ext.colsum_gemv_acc(Kl, yc, Ky)

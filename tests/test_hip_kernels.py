@@ -450,3 +450,30 @@ def test_gpc_fp32_accuracy_tracks_fp64_systematically(dev, ext):
     assert acc_cpu > 0.97, acc_cpu
     assert acc_gpu > 0.97, acc_gpu
     assert abs(acc_gpu - acc_cpu) < 0.01, (acc_gpu, acc_cpu)
+
+
+def test_syrk_sync_matches_plain(dev, ext):
+    """The k-synchronized persistent SYRK must produce the same KK as the
+    split-k kernel (pacing is best-effort and carries no data dependency;
+    the numerics are the identical MFMA chain in a different block order)."""
+    from spark_gp_amd.ops import hip_backend as hb
+    torch.manual_seed(1)
+    m, rows = 4352, 8192          # > 4096 gate; non-multiple-of-256 edge
+    V = torch.rand(rows, m, device=dev, dtype=torch.float32)
+    Kc = V.to(torch.bfloat16)
+    Kl = (V - Kc.float()).to(torch.bfloat16)
+    KcT = Kc.T.contiguous()
+    KlT = Kl.T.contiguous()
+    KK_plain = torch.zeros(m, m, device=dev, dtype=torch.float32)
+    ext.syrk_bf16_acc(KcT, KlT, KK_plain, 8)
+    KK_sync = torch.zeros(m, m, device=dev, dtype=torch.float32)
+    for tt, nact in hb._syrk_sync_tiles(m, dev):
+        ext.syrk_bf16_sync_acc(KcT, KlT, KK_sync, tt, 8, nact)
+    # same products, different accumulation ORDER across split-k slices:
+    # agreement to fp32 rounding of the slice sums
+    ref = (Kc.double() + Kl.double()).T @ (Kc.double() + Kl.double())
+    err_sync = (KK_sync.double() - ref).abs().max().item()
+    err_plain = (KK_plain.double() - ref).abs().max().item()
+    scale = ref.abs().max().item()
+    assert err_sync < 3e-5 * scale, (err_sync, scale)
+    assert err_sync < 4.0 * max(err_plain, 1e-30)
