@@ -73,3 +73,25 @@ if "--sweep" in sys.argv:
         dt = (time.perf_counter() - t0) / 10
         print(f"  split_k={sk:4d}: {dt * 1e3:8.3f} ms  "
               f"{flops / dt / 1e12:6.1f} TF bf16")
+
+# A/B the k-synchronized persistent path (m >= 4096)
+if "--sync" in sys.argv and m >= 4096:
+    from spark_gp_amd.ops import hip_backend as hb
+    launches = hb._syrk_sync_tiles(m, "cuda")
+    print(f"sync: {len(launches)} launches, "
+          f"tiles {[n for _, n in launches]}")
+
+    def run_sync():
+        for tt, nact in launches:
+            ext.syrk_bf16_sync_acc(KcT, KlT, KK, tt, 8, nact)
+
+    for _ in range(3):
+        run_sync()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(10):
+        run_sync()
+    torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / 10
+    print(f"sync: {dt * 1e3:.3f} ms/call  ->  {flops / dt / 1e12:.1f} TF "
+          f"bf16")
