@@ -81,17 +81,17 @@ if "--sync" in sys.argv and m >= 4096:
     print(f"sync: {len(launches)} launches, "
           f"tiles {[n for _, n in launches]}")
 
-    def run_sync():
-        for tt, nact in launches:
-            ext.syrk_bf16_sync_acc(KcT, KlT, KK, tt, 8, nact)
-
-    for _ in range(3):
-        run_sync()
-    torch.cuda.synchronize()
-    t0 = time.perf_counter()
-    for _ in range(10):
-        run_sync()
-    torch.cuda.synchronize()
-    dt = (time.perf_counter() - t0) / 10
-    print(f"sync: {dt * 1e3:.3f} ms/call  ->  {flops / dt / 1e12:.1f} TF "
-          f"bf16")
+    for kpb in (2, 4, 8, 16, 32):
+        def run_sync():
+            for tt, nact in launches:
+                ext.syrk_bf16_sync_acc(KcT, KlT, KK, tt, kpb, nact)
+        for _ in range(2):
+            run_sync()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(8):
+            run_sync()
+        torch.cuda.synchronize()
+        dt = (time.perf_counter() - t0) / 8
+        print(f"sync kpb={kpb:3d}: {dt * 1e3:8.3f} ms/call  ->  "
+              f"{flops / dt / 1e12:6.1f} TF bf16")
