@@ -81,7 +81,7 @@ if "--sync" in sys.argv and m >= 4096:
     print(f"sync: {len(launches)} launches, "
           f"tiles {[n for _, n in launches]}")
 
-    for kpb in (2, 4, 8, 16, 32):
+    for kpb in (8, 32, 64, 128, 100000):
         def run_sync():
             for tt, nact in launches:
                 ext.syrk_bf16_sync_acc(KcT, KlT, KK, tt, kpb, nact)
