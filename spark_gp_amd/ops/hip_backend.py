@@ -232,12 +232,16 @@ def _syrk_sync_tiles(m: int, device):
 
 def _syrk_dispatch(KcT, KlT, KK, m: int):
     import os
-    if m >= 4096 and os.environ.get("SPARK_GP_AMD_SYRK_SYNC", "1") == "1":
-        for tt, nact in _syrk_sync_tiles(m, KK.device):
-            ext.syrk_bf16_sync_acc(KcT, KlT, KK, tt, 8, nact)
-        return
     ntile = (m + 255) // 256
     tiles = ntile * (ntile + 1) // 2
+    # k-synchronized persistent path wins once the tile set exceeds the
+    # resident-block set (measured: m=8192 627 vs 494 TF at kpb=128;
+    # m=4096 the split-k kernel is still ahead, 526 vs ~494 —
+    # PROFILES.md round-2 SYRK section)
+    if tiles > 256 and os.environ.get("SPARK_GP_AMD_SYRK_SYNC", "1") == "1":
+        for tt, nact in _syrk_sync_tiles(m, KK.device):
+            ext.syrk_bf16_sync_acc(KcT, KlT, KK, tt, 128, nact)
+        return
     # split_k: measured sweep (scripts/bench_syrk.py --sweep, r2) — see
     # PROFILES.md; shorter per-block k-ranges keep the drifting column
     # windows closer to L3 residency
