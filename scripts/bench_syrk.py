@@ -76,12 +76,13 @@ if "--sweep" in sys.argv:
 
 # A/B the k-synchronized persistent path (m >= 4096)
 if "--sync" in sys.argv and m >= 4096:
+    import os as _os
     from spark_gp_amd.ops import hip_backend as hb
     launches = hb._syrk_sync_tiles(m, "cuda")
-    print(f"sync: {len(launches)} launches, "
-          f"tiles {[n for _, n in launches]}")
+    print(f"sync patch={_os.environ.get('SPARK_GP_AMD_SYRK_PATCH','2x4')}: "
+          f"{len(launches)} launches, tiles {[n for _, n in launches]}")
 
-    for kpb in (8, 32, 64, 128, 100000):
+    for kpb in (128,):
         def run_sync():
             for tt, nact in launches:
                 ext.syrk_bf16_sync_acc(KcT, KlT, KK, tt, kpb, nact)

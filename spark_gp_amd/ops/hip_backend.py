@@ -208,12 +208,15 @@ def _syrk_sync_tiles(m: int, device):
     tiles are sorted by 2x4-tile patches — an XCD's resident tiles then
     share a handful of 256-row/column operand windows that fit its 4 MB L2
     at the 256-k phase width."""
-    key = (m, str(device))
+    import os
+    pr, pc = (int(v) for v in
+              os.environ.get("SPARK_GP_AMD_SYRK_PATCH", "2x4").split("x"))
+    key = (m, str(device), pr, pc)
     if key in _SYNC_TILE_CACHE:
         return _SYNC_TILE_CACHE[key]
     ntile = (m + 255) // 256
     tiles = [(ti, tj) for ti in range(ntile) for tj in range(ti + 1)]
-    tiles.sort(key=lambda t: (t[0] // 2, t[1] // 4))
+    tiles.sort(key=lambda t: (t[0] // pr, t[1] // pc))
     launches = []
     CAP = 224
     for s in range(0, len(tiles), CAP):
