@@ -74,10 +74,28 @@ class Comm:
             t.copy_(buf.to(t.device))
         return t
 
+    # persistent staging buffer for host-side payloads under the nccl
+    # backend: avoids a fresh allocation + H2D/D2H pair per objective
+    # evaluation (the (1+p) C1 allreduce fires once per L-BFGS eval)
+    _np_stage: Optional[torch.Tensor] = None
+
     def allreduce_np(self, arr: np.ndarray, op: str = "sum") -> np.ndarray:
         if not self._active:
             return arr
         t = torch.from_numpy(np.ascontiguousarray(arr))
+        if dist.get_backend() == "nccl":
+            cls = type(self)
+            if (cls._np_stage is None or cls._np_stage.numel() < t.numel()
+                    or cls._np_stage.dtype != t.dtype):
+                cls._np_stage = torch.empty(t.numel(), dtype=t.dtype,
+                                            device="cuda")
+            buf = cls._np_stage[:t.numel()].view_as(t)
+            buf.copy_(t, non_blocking=False)
+            red = {"sum": dist.ReduceOp.SUM, "min": dist.ReduceOp.MIN,
+                   "max": dist.ReduceOp.MAX}[op]
+            dist.all_reduce(buf, op=red)
+            t.copy_(buf.cpu())
+            return t.numpy()
         self.allreduce_(t, op)
         return t.numpy()
 
