@@ -146,3 +146,18 @@ def test_dpotrf64_big_m():
     x = hb.chol_solve64(Lp, V, b, m)
     err = (M @ x - b).abs().max().item() / b.abs().max().item()
     assert err < 1e-9, err
+
+
+@pytest.mark.parametrize("m", [1, 63, 64, 65])
+def test_dpotrf64_boundary_sizes(m):
+    """Padding boundaries: m below/at/above one 64-block."""
+    from spark_gp_amd.ops import hip_backend as hb
+    M = _spd(m, seed=m + 40)
+    Lp, V = hb.chol_factor64(M)
+    L_ref = torch.linalg.cholesky(M)
+    torch.testing.assert_close(torch.tril(Lp[:m, :m]), L_ref,
+                               rtol=1e-11, atol=1e-9)
+    b = torch.randn(m, 2, dtype=torch.float64).cuda()
+    X = hb.chol_solve64(Lp, V, b, m)
+    torch.testing.assert_close(X, torch.cholesky_solve(b, L_ref),
+                               rtol=1e-9, atol=1e-9)

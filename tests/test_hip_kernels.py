@@ -498,3 +498,23 @@ def test_greedy_provider_gpu_path(dev, ext):
     pred = model.predict(X[:5000])
     rmse = float(np.sqrt(np.mean((pred - y[:5000]) ** 2)))
     assert rmse < 0.15, rmse
+
+
+def test_probit_classifier_gpu(dev, ext):
+    """Config-3 path on GPU: probit link (torch fp32 Laplace — the fused
+    kernel covers logistic; probit must still run end to end on device)."""
+    from spark_gp_amd import GaussianProcessClassifier
+    from spark_gp_amd.kernels import RBFKernel
+    from spark_gp_amd.data import mnist_like_binary
+    from spark_gp_amd.utils.scaling import StandardScaler
+    X, y = mnist_like_binary(2400, seed=13)
+    Xs = StandardScaler().fit_transform(X)
+    model = (GaussianProcessClassifier()
+             .setKernel(lambda: 1 * RBFKernel(10.0))
+             .setLink("probit")
+             .setDatasetSizeForExpert(100).setActiveSetSize(300)
+             .setSigma2(1e-3).setTol(1e-3).setMaxIter(30).setSeed(13)
+             .setDevice("cuda")
+             .fit(Xs, y))
+    acc = float((model.predict(Xs) == y).mean())
+    assert acc > 0.97, acc
