@@ -334,3 +334,50 @@ def test_distributed_poisson_agrees_across_ranks():
     true_rate = np.exp(1.0 + np.sin(3 * X.sum(-1)))
     rel = np.abs(res[0] - true_rate[:40]) / true_rate[:40]
     assert np.median(rel) < 0.35, np.median(rel)
+
+
+# ---- world_size=3, uneven shards (scaling-bench shape insurance) ----------
+
+def _run3(rank, world_size, port, out_q):
+    import torch.distributed as dist
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world_size)
+    try:
+        from spark_gp_amd import GaussianProcessRegression
+        from spark_gp_amd.kernels import ARDRBFKernel, EyeKernel, Scalar
+        X, y = performance_benchmark_data(601, 3, seed=9, dtype=np.float64)
+        y = np.sin(X.sum(-1) * 3.0)
+        Xl, yl = _shard(X, y, rank, world_size)   # 201/200/200 rows
+        model = (GaussianProcessRegression()
+                 .setKernel(lambda: 1 * ARDRBFKernel(3)
+                            + Scalar(1e-2).const * EyeKernel())
+                 .setDatasetSizeForExpert(50).setActiveSetSize(60)
+                 .setSigma2(1e-3).setMaxIter(30).setSeed(5).setDevice("cpu")
+                 .fit(Xl, yl))
+        out_q.put((rank, model.predict(X[:40])))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_three_rank_uneven_shards():
+    """world_size=3 with a 201/200/200 split: the N>2 collective paths the
+    driver's 4- and 8-GPU scaling bench will exercise (global sampling
+    offsets, objective sum, PPA stats) must give identical models."""
+    ctx = mp.get_context("spawn")
+    out_q = ctx.Queue()
+    procs = [ctx.Process(target=_run3, args=(r, 3, 29871, out_q))
+             for r in range(3)]
+    for p in procs:
+        p.start()
+    res = {}
+    for _ in range(3):
+        rank, pred = out_q.get(timeout=300)
+        res[rank] = pred
+    for p in procs:
+        p.join(timeout=60)
+    np.testing.assert_allclose(res[0], res[1], atol=1e-12)
+    np.testing.assert_allclose(res[0], res[2], atol=1e-12)
+    X, _ = performance_benchmark_data(601, 3, seed=9, dtype=np.float64)
+    yq = np.sin(X[:40].sum(-1) * 3.0)
+    assert float(np.sqrt(np.mean((res[0] - yq) ** 2))) < 0.2
