@@ -221,6 +221,11 @@ __device__ inline void chol_invert_lower(float* Abuf, float* Tbuf,
       const int qb = q * 8;
       const int sbs = min(8, bs - qb);
       if (tid < 64) {
+        // static priority for the serial factor wave: its dependent
+        // shfl/VALU chain is the block's critical path while waves 1-7
+        // stream trailing updates — give it the issue-arbitration edge
+        // (MI355X_MICROARCH.md "Two waves per SIMD", item 4)
+        __builtin_amdgcn_s_setprio(1);
         {
           // 8x8 factor + trtri on lanes 0..7, ROW j per lane, cross-lane
           // traffic via __shfl (ds_bpermute).  Row-per-lane keeps every
@@ -312,6 +317,7 @@ __device__ inline void chol_invert_lower(float* Abuf, float* Tbuf,
             }
           }
         }
+        __builtin_amdgcn_s_setprio(0);
       } else if (J > 0) {
         // waves 1-7, chunk q (strided by nq, so any nq covers everything):
         // previous panel's update to (a) the panel rows of column-block J
