@@ -34,7 +34,7 @@ ext = CUDAExtension(
 
 setup(
     name="spark_gp_amd",
-    version="0.1.0",
+    version="0.2.0",
     packages=find_packages(include=["spark_gp_amd", "spark_gp_amd.*"]),
     package_data={"spark_gp_amd": ["data/*.csv", "ops/csrc/*"]},
     ext_modules=[ext],

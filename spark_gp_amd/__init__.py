@@ -36,7 +36,7 @@ from .ppa import NotPositiveDefiniteError
 from .utils import (Integrator, OneVsRest, StandardScaler, accuracy,
                     cross_validate, rmse, scale, train_validation_split)
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
 
 __all__ = [
     "GaussianProcessRegression", "GaussianProcessRegressionModel",
