@@ -46,6 +46,7 @@
 
 #include <hip/hip_runtime.h>
 #include <math.h>
+#include <stdlib.h>
 
 #define WG 512
 #include "linalg_lds.h"
@@ -331,6 +332,10 @@ extern "C" hipError_t launch_fused_expert_nll(
     int* out_bad, unsigned long long* out_clk, hipStream_t stream,
     size_t* lds_used) {
   size_t lds = nll_lds_bytes2(k, d);
+  // occupancy experiment knob: force 1 WG/CU by padding the LDS ask
+  // (measures how much the usual 2 co-resident expert WGs overlap)
+  if (const char* pad = getenv("SPARK_GP_AMD_NLL_LDS_PAD"))
+    lds += (size_t)atol(pad);
   if (lds_used) *lds_used = lds;
   if (lds > 160 * 1024 || k > 128 || d > 128)
     return hipErrorInvalidConfiguration;
