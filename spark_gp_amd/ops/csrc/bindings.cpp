@@ -379,8 +379,13 @@ void dpotrf64(torch::Tensor A, torch::Tensor V, torch::Tensor bad) {
 }
 
 // Solve (L L^T) X = B in place on B [mp, r] given the factored A (lower L)
-// and the diagonal-block inverses V from dpotrf64.
-void dchol_solve64(torch::Tensor A, torch::Tensor V, torch::Tensor B) {
+// and the diagonal-block inverses V from dpotrf64.  rhs_identity=true
+// declares B to be the identity (the explicit-inverse path): the forward
+// pass then skips the provably-zero column range of each block step —
+// Y = L^{-1} is lower triangular, so block row I has nonzeros only in
+// columns < (I+1)*64 — cutting the forward GEMMs from m^3 to m^3/3.
+void dchol_solve64(torch::Tensor A, torch::Tensor V, torch::Tensor B,
+                   bool rhs_identity) {
   TORCH_CHECK(A.is_cuda() && A.dtype() == torch::kFloat64 &&
               A.is_contiguous() && A.dim() == 2);
   TORCH_CHECK(B.is_cuda() && B.dtype() == torch::kFloat64 &&
@@ -395,14 +400,15 @@ void dchol_solve64(torch::Tensor A, torch::Tensor V, torch::Tensor B) {
   double* b = B.data_ptr<double>();
   for (long I = 0; I < nb; ++I) {            // forward: L Y = B
     const long jb = I * 64;
+    const long rc = rhs_identity ? std::min<long>(r, jb + 64) : r;
     check_hip(launch_dgemm64(0, 0, 0, 0, v + I * 4096, b + jb * r,
-                             b + jb * r, 64, (int)r, 64, 64, r, r, stream),
+                             b + jb * r, 64, (int)rc, 64, 64, r, r, stream),
               "fwd_diag");
     const long nr = mp - jb - 64;
     if (nr > 0)
       check_hip(launch_dgemm64(0, 0, 1, 0, a + (size_t)(jb + 64) * mp + jb,
                                b + jb * r, b + (jb + 64) * r,
-                               (int)nr, (int)r, 64, mp, r, r, stream),
+                               (int)nr, (int)rc, 64, mp, r, r, stream),
                 "fwd_update");
   }
   for (long I = nb - 1; I >= 0; --I) {       // backward: L^T X = Y
@@ -457,7 +463,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("dpotrf64", &dpotrf64,
           "blocked fp64 Cholesky (K13), in place, 64-padded");
   mod.def("dchol_solve64", &dchol_solve64,
-          "blocked fp64 triangular solves (L L^T) X = B, in place on B");
+          "blocked fp64 triangular solves (L L^T) X = B, in place on B",
+          pybind11::arg("A"), pybind11::arg("V"), pybind11::arg("B"),
+          pybind11::arg("rhs_identity") = false);
   mod.def("dgemm64", &dgemm64, "fp64 MFMA GEMM (test/utility)");
   mod.def("fused_laplace_newton", &fused_laplace_newton,
           "per-expert Laplace Newton loop to convergence (CDNA4)");

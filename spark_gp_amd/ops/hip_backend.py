@@ -361,7 +361,7 @@ def chol_solve64(L: torch.Tensor, V: torch.Tensor, B: torch.Tensor,
 def chol_inverse64(L: torch.Tensor, V: torch.Tensor, m: int) -> torch.Tensor:
     mp = L.shape[0]
     B = torch.eye(mp, dtype=torch.float64, device=L.device)
-    ext.dchol_solve64(L, V, B)
+    ext.dchol_solve64(L, V, B, rhs_identity=True)
     return B[:m, :m]
 
 
