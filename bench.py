@@ -40,7 +40,7 @@ def parse_args():
                    help="L-BFGS-B iteration cap per fit (the reference "
                         "default; measured convergence on the 10M x 32 "
                         "target is ~15 iterations / 19 objective evals, so "
-                        "the cap is not binding — gpurun_out/bench_r2_trace"
+                        "the cap is not binding — profiles/bench_r2_maxiter_trace"
                         ".log / BASELINE.md)")
     p.add_argument("--sigma2", type=float, default=1e-3)
     p.add_argument("--seed", type=int, default=13)
