@@ -15,7 +15,7 @@ extern "C" hipError_t launch_fused_expert_nll(
 extern "C" hipError_t launch_cross_kernel_tile(
     const float* X, const float* A, const float* s2v, float amp, int c,
     int m, int d, void* out, void* out_lo, void* out_t, void* out_lo_t,
-    int out_is_bf16, hipStream_t stream);
+    int out_is_bf16, const float* yv, double* Ky, hipStream_t stream);
 
 extern "C" hipError_t launch_syrk_bf16(const void* KcT, const void* KlT,
                                        int c, int m, int cpitch, int split_k,
@@ -168,11 +168,46 @@ std::vector<torch::Tensor> cross_kernel_tile(torch::Tensor X, torch::Tensor A,
                                      hilo ? lo.data_ptr() : nullptr,
                                      want_t ? outT.data_ptr() : nullptr,
                                      want_t ? loT.data_ptr() : nullptr,
-                                     bf16_out ? 1 : 0, current_stream()),
+                                     bf16_out ? 1 : 0, nullptr, nullptr,
+                                     current_stream()),
             "cross_kernel_tile");
   if (want_t) return {out, lo, outT, loT};
   if (hilo) return {out, lo};
   return {out};
+}
+
+// PPA fast path: transposed hi/lo tiles ONLY (no [c, m] copies written)
+// plus the fused column-sum Ky += K^T y accumulated from the fp32
+// register values (replaces the colsum_gemv re-read pass).
+std::vector<torch::Tensor> cross_kernel_tile_ppa(torch::Tensor X,
+                                                 torch::Tensor A,
+                                                 torch::Tensor s2v,
+                                                 double amp, torch::Tensor y,
+                                                 torch::Tensor Ky) {
+  TORCH_CHECK(X.is_cuda() && X.dtype() == torch::kFloat32 && X.dim() == 2);
+  TORCH_CHECK(A.is_cuda() && A.dtype() == torch::kFloat32 && A.dim() == 2);
+  TORCH_CHECK(X.size(1) == A.size(1), "feature dims differ");
+  TORCH_CHECK(y.is_cuda() && y.dtype() == torch::kFloat32 &&
+              y.numel() == X.size(0));
+  TORCH_CHECK(Ky.is_cuda() && Ky.dtype() == torch::kFloat64 &&
+              Ky.numel() == A.size(0));
+  auto Xc = X.contiguous();
+  auto Ac = A.contiguous();
+  auto sc = s2v.contiguous();
+  auto yc = y.contiguous();
+  const int c = X.size(0), m = A.size(0), d = X.size(1);
+  auto opts = torch::TensorOptions().dtype(torch::kBFloat16)
+                  .device(X.device());
+  auto outT = torch::empty({m, c}, opts);
+  auto loT = torch::empty({m, c}, opts);
+  check_hip(launch_cross_kernel_tile(Xc.data_ptr<float>(), Ac.data_ptr<float>(),
+                                     sc.data_ptr<float>(), (float)amp, c, m, d,
+                                     nullptr, nullptr, outT.data_ptr(),
+                                     loT.data_ptr(), 1,
+                                     yc.data_ptr<float>(),
+                                     Ky.data_ptr<double>(), current_stream()),
+            "cross_kernel_tile_ppa");
+  return {outT, loT};
 }
 
 // KcT/KlT are the TRANSPOSED kernel blocks [m, c] (k along the contiguous
@@ -479,6 +514,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("fused_expert_nll_profile", &fused_expert_nll_impl,
           "same, with per-phase wall_clock64 boundaries appended");
   mod.def("fused_expert_nll_supported", &fused_expert_nll_supported);
+  mod.def("cross_kernel_tile_ppa", &cross_kernel_tile_ppa,
+          "transposed hi/lo tiles + fused K^T y accumulation (CDNA4)");
   mod.def("cross_kernel_tile", &cross_kernel_tile,
           "rectangular RBF/ARD kernel block (CDNA4)",
           pybind11::arg("X"), pybind11::arg("A"), pybind11::arg("s2v"),

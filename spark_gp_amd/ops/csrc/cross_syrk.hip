@@ -39,10 +39,13 @@ cross_kernel_tile_kernel(const float* __restrict__ X,    // [c, d]
                                                          // residual v - hi
                          bf16* __restrict__ out_bfT,     // [m, c] or null
                          bf16* __restrict__ out_loT,     // [m, c] or null
-                         float* __restrict__ out_f32) {  // [c, m] or null
+                         float* __restrict__ out_f32,    // [c, m] or null
+                         const float* __restrict__ yv,   // [c] or null
+                         double* __restrict__ Ky) {      // [m] (with yv)
   __shared__ float xs[CK_TILE][CK_DBLK + 1];
   __shared__ float as[CK_TILE][CK_DBLK + 1];
   __shared__ float s2s[CK_DBLK];
+  __shared__ double kyred[16][16 + 1];    // fused colsum partials
 
   const int row0 = blockIdx.x * CK_TILE;
   const int col0 = blockIdx.y * CK_TILE;
@@ -140,6 +143,40 @@ cross_kernel_tile_kernel(const float* __restrict__ X,    // [c, d]
           out_bfT[(size_t)gc * c + gr0 + i] = th[i];
           if (out_loT) out_loT[(size_t)gc * c + gr0 + i] = tl[i];
         }
+      }
+    }
+  }
+
+  if (yv) {
+    // fused colsum (K12's K_mn^T y): accumulate this tile's column sums
+    // of K * y straight from the fp32 register values — the separate
+    // colsum pass used to RE-READ the whole [c, m] block from HBM (and
+    // was the only consumer of the non-transposed copies)
+    float s[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float sj = 0.f;
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        const int gr = row0 + tr + i;
+        const float yw = (gr < c) ? yv[gr] : 0.f;
+        sj += acc[i][j] * yw;
+      }
+      s[j] = sj;
+    }
+    const int thr = threadIdx.x >> 4;       // tr / 8 in thread units
+    const int thc = threadIdx.x & 15;       // tc / 8 in thread units
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      __syncthreads();
+      kyred[thr][thc] = (double)s[j];
+      __syncthreads();
+      if (thr == 0) {
+        double t = 0.0;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) t += kyred[r][thc];
+        const int gc = col0 + thc * 8 + j;
+        if (gc < m) atomicAdd(&Ky[gc], t);
       }
     }
   }
@@ -524,14 +561,15 @@ colsum_gemv_kernel(const bf16* __restrict__ Kc,   // [c, m]
 extern "C" hipError_t launch_cross_kernel_tile(
     const float* X, const float* A, const float* s2v, float amp,
     int c, int m, int d, void* out, void* out_lo, void* out_t,
-    void* out_lo_t, int out_is_bf16, hipStream_t stream) {
+    void* out_lo_t, int out_is_bf16, const float* yv, double* Ky,
+    hipStream_t stream) {
   dim3 grid((c + CK_TILE - 1) / CK_TILE, (m + CK_TILE - 1) / CK_TILE);
   hipLaunchKernelGGL(cross_kernel_tile_kernel, grid, dim3(256), 0, stream,
                      X, A, s2v, amp, c, m, d,
                      out_is_bf16 ? (bf16*)out : nullptr,
                      out_is_bf16 ? (bf16*)out_lo : nullptr,
                      (bf16*)out_t, (bf16*)out_lo_t,
-                     out_is_bf16 ? nullptr : (float*)out);
+                     out_is_bf16 ? nullptr : (float*)out, yv, Ky);
   return hipGetLastError();
 }
 

@@ -291,15 +291,16 @@ def kmn_knm_and_kmny(kernel: Kernel, active: torch.Tensor,
     KK = torch.zeros(m, m, dtype=torch.float32, device=X.device)
     for s in range(0, n, chunk_rows):
         e = min(n, s + chunk_rows)
-        # the cross kernel also emits transposed (k-contiguous) copies so
-        # the MFMA SYRK stages with full-line loads and b128 LDS traffic
-        Kc, Kl, KcT, KlT = ext.cross_kernel_tile(X[s:e].contiguous(), act32,
-                                                 s2, float(C), True, True,
-                                                 True)
+        # fast PPA tile: the cross kernel writes ONLY the transposed
+        # (k-contiguous) hi/lo copies the MFMA SYRK stages from, and
+        # accumulates Ky += K^T y from its fp32 register values in the
+        # same launch (the old separate colsum pass re-read the whole
+        # [c, m] block from HBM and was the only consumer of the
+        # non-transposed copies)
+        KcT, KlT = ext.cross_kernel_tile_ppa(X[s:e].contiguous(), act32,
+                                             s2, float(C),
+                                             y32[s:e].contiguous(), Ky)
         _syrk_dispatch(KcT, KlT, KK, m)
-        yc = y32[s:e].contiguous()
-        ext.colsum_gemv_acc(Kc, yc, Ky)
-        ext.colsum_gemv_acc(Kl, yc, Ky)
     return KK.double(), Ky
 
 
