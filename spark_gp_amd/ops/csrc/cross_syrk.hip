@@ -115,7 +115,7 @@ cross_kernel_tile_kernel(const float* __restrict__ X,    // [c, d]
       if (out_bf) {
         out_bf[(size_t)gr * m + gc] = hv[i][j];
         if (out_lo) out_lo[(size_t)gr * m + gc] = lv[i][j];
-      } else {
+      } else if (out_f32) {
         out_f32[(size_t)gr * m + gc] = acc[i][j];
       }
     }
