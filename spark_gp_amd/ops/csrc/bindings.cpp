@@ -28,6 +28,13 @@ extern "C" hipError_t launch_syrk_bf16_sync(const void* KcT, const void* KlT,
                                             int nactive, float* KK,
                                             hipStream_t stream);
 
+extern "C" hipError_t launch_cross_mfma(const float* Xs, const float* As,
+                                         const float* nx, const float* na,
+                                         float amp, int c, int m, int d,
+                                         void* out_bfT, void* out_loT,
+                                         const float* yv, double* Ky,
+                                         hipStream_t stream);
+
 extern "C" hipError_t launch_colsum_gemv(const void* Kc, const float* y,
                                          int c, int m, double* Ky,
                                          hipStream_t stream);
@@ -174,6 +181,36 @@ std::vector<torch::Tensor> cross_kernel_tile(torch::Tensor X, torch::Tensor A,
   if (want_t) return {out, lo, outT, loT};
   if (hilo) return {out, lo};
   return {out};
+}
+
+// MFMA PPA tile: pre-scaled inputs + norms; K1's sqdist-via-GEMM plan.
+std::vector<torch::Tensor> cross_mfma_ppa(torch::Tensor Xs, torch::Tensor As,
+                                          torch::Tensor nx, torch::Tensor na,
+                                          double amp, torch::Tensor y,
+                                          torch::Tensor Ky) {
+  TORCH_CHECK(Xs.is_cuda() && Xs.dtype() == torch::kFloat32 && Xs.dim() == 2);
+  TORCH_CHECK(As.is_cuda() && As.dtype() == torch::kFloat32 && As.dim() == 2);
+  TORCH_CHECK(Xs.size(1) == As.size(1));
+  const int c = Xs.size(0), m = As.size(0), d = Xs.size(1);
+  TORCH_CHECK(nx.numel() == c && na.numel() == m && y.numel() == c);
+  TORCH_CHECK(Ky.is_cuda() && Ky.dtype() == torch::kFloat64 &&
+              Ky.numel() == m);
+  auto Xc = Xs.contiguous();
+  auto Ac = As.contiguous();
+  auto nxc = nx.contiguous();
+  auto nac = na.contiguous();
+  auto yc = y.contiguous();
+  auto opts = torch::TensorOptions().dtype(torch::kBFloat16)
+                  .device(Xs.device());
+  auto outT = torch::empty({m, c}, opts);
+  auto loT = torch::empty({m, c}, opts);
+  check_hip(launch_cross_mfma(Xc.data_ptr<float>(), Ac.data_ptr<float>(),
+                              nxc.data_ptr<float>(), nac.data_ptr<float>(),
+                              (float)amp, c, m, d, outT.data_ptr(),
+                              loT.data_ptr(), yc.data_ptr<float>(),
+                              Ky.data_ptr<double>(), current_stream()),
+            "cross_mfma_ppa");
+  return {outT, loT};
 }
 
 // PPA fast path: transposed hi/lo tiles ONLY (no [c, m] copies written)
@@ -514,6 +551,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("fused_expert_nll_profile", &fused_expert_nll_impl,
           "same, with per-phase wall_clock64 boundaries appended");
   mod.def("fused_expert_nll_supported", &fused_expert_nll_supported);
+  mod.def("cross_mfma_ppa", &cross_mfma_ppa,
+          "MFMA sqdist cross tile + fused K^T y (K1 plan, CDNA4)");
   mod.def("cross_kernel_tile_ppa", &cross_kernel_tile_ppa,
           "transposed hi/lo tiles + fused K^T y accumulation (CDNA4)");
   mod.def("cross_kernel_tile", &cross_kernel_tile,

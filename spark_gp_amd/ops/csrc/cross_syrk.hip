@@ -183,6 +183,137 @@ cross_kernel_tile_kernel(const float* __restrict__ X,    // [c, d]
 }
 
 // ---------------------------------------------------------------------------
+// cross_mfma: the K1 MFMA plan — sqdist via ||x||^2 + ||a||^2 - 2 x.a on
+// f32 matrix cores (v_mfma_f32_16x16x4_f32: exact fp32 at 157 TF, vs the
+// elementwise kernel's ~3 VALU issues per (element, dim)).  PPA mode only:
+// inputs arrive PRE-SCALED (x' = x * s per dim) with per-row squared norms;
+// outputs are the transposed hi/lo bf16 tiles + the fused Ky += K^T y.
+// ---------------------------------------------------------------------------
+
+#define CM_TILE 128
+#define CM_DBLK 32
+#define CM_STR 36           // f32 LDS row stride ([i][k] layout)
+
+typedef __attribute__((ext_vector_type(4))) float f32x4_;
+
+extern "C" __global__ void __launch_bounds__(256)
+cross_mfma_kernel(const float* __restrict__ Xs,   // [c, d] pre-scaled
+                  const float* __restrict__ As,   // [m, d] pre-scaled
+                  const float* __restrict__ nx,   // [c] ||x'||^2
+                  const float* __restrict__ na,   // [m] ||a'||^2
+                  const float amp,
+                  const int c, const int m, const int d,
+                  bf16* __restrict__ out_bfT,     // [m, c]
+                  bf16* __restrict__ out_loT,     // [m, c]
+                  const float* __restrict__ yv,   // [c]
+                  double* __restrict__ Ky) {      // [m]
+  __shared__ float xt[CM_TILE * CM_STR];
+  __shared__ float at[CM_TILE * CM_STR];
+  __shared__ float kyd[CM_TILE];
+
+  const int row0 = blockIdx.x * CM_TILE;          // x rows
+  const int col0 = blockIdx.y * CM_TILE;          // a rows (K columns)
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;                      // 4 waves, 2x2 grid
+  const int lane = tid & 63;
+  const int wr = (wave >> 1) * 64;
+  const int wc = (wave & 1) * 64;
+  const int l16 = lane & 15, kg = lane >> 4;
+
+  for (int f = tid; f < CM_TILE; f += 256) kyd[f] = 0.f;
+
+  // 4x4 tiles of 16x16 per wave = a 64x64 sub-tile
+  f32x4_ acc[4][4];
+#pragma unroll
+  for (int a = 0; a < 4; ++a)
+#pragma unroll
+    for (int b = 0; b < 4; ++b) acc[a][b] = {0.f, 0.f, 0.f, 0.f};
+
+  for (int d0 = 0; d0 < d; d0 += CM_DBLK) {
+    const int dl = min(CM_DBLK, d - d0);
+    for (int f = tid; f < CM_TILE * CM_DBLK; f += 256) {
+      const int r = f >> 5, q = f & 31;           // [i][k], coalesced on k
+      const int gr = row0 + r, gc = col0 + r;
+      xt[r * CM_STR + q] = (gr < c && q < dl)
+                               ? Xs[(size_t)gr * d + d0 + q] : 0.f;
+      at[r * CM_STR + q] = (gc < m && q < dl)
+                               ? As[(size_t)gc * d + d0 + q] : 0.f;
+    }
+    __syncthreads();
+#pragma unroll
+    for (int kc = 0; kc < CM_DBLK / 4; ++kc) {
+      const int fk = kg + kc * 4;
+      float xa[4], ab[4];
+#pragma unroll
+      for (int a = 0; a < 4; ++a)
+        xa[a] = xt[(wr + a * 16 + l16) * CM_STR + fk];
+#pragma unroll
+      for (int b = 0; b < 4; ++b)
+        ab[b] = at[(wc + b * 16 + l16) * CM_STR + fk];
+#pragma unroll
+      for (int a = 0; a < 4; ++a)
+#pragma unroll
+        for (int b = 0; b < 4; ++b)
+          acc[a][b] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+              xa[a], ab[b], acc[a][b], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // epilogue: q = nx + na - 2 dot; v = amp exp(-q); hi/lo bf16; stores
+  // transposed (4 consecutive x-rows per fragment reg -> one 8-B store);
+  // fused Ky partials through LDS.
+  // f32 16x16x4 D map: col = lane&15, row = (lane>>4)*4 + reg (the
+  // dtype-independent standard map — unlike the f64 form, which was
+  // measured transposed in reg; verified by the parity unit test).
+  const int crow = (lane >> 4) * 4;
+  const int ccol = l16;
+#pragma unroll
+  for (int a = 0; a < 4; ++a) {
+    const int gr0 = row0 + wr + a * 16 + crow;
+#pragma unroll
+    for (int b = 0; b < 4; ++b) {
+      const int gc = col0 + wc + b * 16 + ccol;
+      if (gc >= m) continue;
+      const float nav = na[gc];
+      bf16 th[4], tl[4];
+      float kysum = 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int gr = gr0 + r;
+        float v = 0.f;
+        if (gr < c) {
+          const float q = nx[gr] + nav - 2.0f * acc[a][b][r];
+          v = amp * __expf(-(q > 0.f ? q : 0.f));
+          kysum += v * yv[gr];
+        }
+        const bf16 h = (bf16)v;
+        th[r] = h;
+        tl[r] = (bf16)(v - (float)h);
+      }
+      if ((c % 4 == 0) && gr0 + 4 <= c) {   // 8-B alignment needs 4 | c
+        *(uint2*)&out_bfT[(size_t)gc * c + gr0] = *(uint2*)th;
+        *(uint2*)&out_loT[(size_t)gc * c + gr0] = *(uint2*)tl;
+      } else {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          if (gr0 + r < c) {
+            out_bfT[(size_t)gc * c + gr0 + r] = th[r];
+            out_loT[(size_t)gc * c + gr0 + r] = tl[r];
+          }
+        }
+      }
+      atomicAdd(&kyd[wc + b * 16 + ccol], kysum);
+    }
+  }
+  __syncthreads();
+  for (int f = tid; f < CM_TILE; f += 256) {
+    const int gc = col0 + f;
+    if (gc < m && kyd[f] != 0.f) atomicAdd(&Ky[gc], (double)kyd[f]);
+  }
+}
+
+// ---------------------------------------------------------------------------
 // syrk_bf16: KK[m, m] += K^T K from the TRANSPOSED chunk KcT [m, c]
 // ---------------------------------------------------------------------------
 // 512-thread block = 8 waves in a 4x2 grid; 256x256 output tile per block
@@ -570,6 +701,19 @@ extern "C" hipError_t launch_cross_kernel_tile(
                      out_is_bf16 ? (bf16*)out_lo : nullptr,
                      (bf16*)out_t, (bf16*)out_lo_t,
                      out_is_bf16 ? nullptr : (float*)out, yv, Ky);
+  return hipGetLastError();
+}
+
+extern "C" hipError_t launch_cross_mfma(const float* Xs, const float* As,
+                                         const float* nx, const float* na,
+                                         float amp, int c, int m, int d,
+                                         void* out_bfT, void* out_loT,
+                                         const float* yv, double* Ky,
+                                         hipStream_t stream) {
+  dim3 grid((c + CM_TILE - 1) / CM_TILE, (m + CM_TILE - 1) / CM_TILE);
+  hipLaunchKernelGGL(cross_mfma_kernel, grid, dim3(256), 0, stream,
+                     Xs, As, nx, na, amp, c, m, d,
+                     (bf16*)out_bfT, (bf16*)out_loT, yv, Ky);
   return hipGetLastError();
 }
 

@@ -289,17 +289,28 @@ def kmn_knm_and_kmny(kernel: Kernel, active: torch.Tensor,
     # MFMA SYRK keeps input-quantization error at fp32 class while running
     # at bf16 matrix-core rate
     KK = torch.zeros(m, m, dtype=torch.float32, device=X.device)
+    # MFMA cross tile (K1 plan): sqdist via ||x'||^2 + ||a'||^2 - 2 x'.a'
+    # on f32 matrix cores with pre-scaled coordinates, writing ONLY the
+    # transposed hi/lo copies the SYRK stages from and accumulating
+    # Ky += K^T y in the same launch.  The elementwise kernel (~3 VALU
+    # issues per (element, dim)) remains as the fallback/AB path.
+    import os
+    use_mfma = os.environ.get("SPARK_GP_AMD_CROSS_MFMA", "1") == "1"
+    if use_mfma:
+        svec = _scale_vector(cs, theta, d, X.device)
+        As = (act32 * svec).contiguous()
+        na = (As * As).sum(-1).contiguous()
     for s in range(0, n, chunk_rows):
         e = min(n, s + chunk_rows)
-        # fast PPA tile: the cross kernel writes ONLY the transposed
-        # (k-contiguous) hi/lo copies the MFMA SYRK stages from, and
-        # accumulates Ky += K^T y from its fp32 register values in the
-        # same launch (the old separate colsum pass re-read the whole
-        # [c, m] block from HBM and was the only consumer of the
-        # non-transposed copies)
-        KcT, KlT = ext.cross_kernel_tile_ppa(X[s:e].contiguous(), act32,
-                                             s2, float(C),
-                                             y32[s:e].contiguous(), Ky)
+        if use_mfma:
+            Xs = (X[s:e] * svec).contiguous()
+            nx = (Xs * Xs).sum(-1).contiguous()
+            KcT, KlT = ext.cross_mfma_ppa(Xs, As, nx, na, float(C),
+                                          y32[s:e].contiguous(), Ky)
+        else:
+            KcT, KlT = ext.cross_kernel_tile_ppa(X[s:e].contiguous(), act32,
+                                                 s2, float(C),
+                                                 y32[s:e].contiguous(), Ky)
         _syrk_dispatch(KcT, KlT, KK, m)
     return KK.double(), Ky
 

@@ -518,3 +518,31 @@ def test_probit_classifier_gpu(dev, ext):
              .fit(Xs, y))
     acc = float((model.predict(Xs) == y).mean())
     assert acc > 0.97, acc
+
+
+def test_cross_mfma_ppa_vs_oracle(dev, ext):
+    """The MFMA sqdist cross tile (K1 plan): hi+lo values and the fused
+    K^T y against the fp64 oracle, odd shapes included."""
+    from spark_gp_amd.kernels import ARDRBFKernel, EyeKernel, Scalar
+    torch.manual_seed(4)
+    for c, m, d in ((1000, 300, 8), (131072 // 16, 1000, 32), (513, 130, 5)):
+        X = torch.rand(c, d, device=dev)
+        A = torch.rand(m, d, device=dev)
+        y = torch.rand(c, device=dev)
+        beta = torch.rand(d, device=dev) + 0.5
+        amp = 1.3
+        Xs = (X * beta).contiguous()
+        As = (A * beta).contiguous()
+        nx = (Xs * Xs).sum(-1).contiguous()
+        na = (As * As).sum(-1).contiguous()
+        Ky = torch.zeros(m, dtype=torch.float64, device=dev)
+        KcT, KlT = ext.cross_mfma_ppa(Xs, As, nx, na, amp, y, Ky)
+        V = KcT.float() + KlT.float()                 # [m, c]
+        kernel = amp * ARDRBFKernel(beta.cpu().numpy())
+        ref = kernel.cross_kernel(X.double().cpu(), A.double().cpu()).T
+        # hi/lo split carries ~16 mantissa bits; the sqdist-via-norms form
+        # adds fp32 cancellation of order eps*||x||^2
+        err = (V.cpu().double() - ref).abs().max().item()
+        assert err < 5e-5 * float(ref.abs().max()), (c, m, d, err)
+        Ky_ref = ref.to(dev) @ y.double()
+        torch.testing.assert_close(Ky, Ky_ref, rtol=1e-4, atol=1e-4)
