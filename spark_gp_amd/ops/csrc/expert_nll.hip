@@ -51,6 +51,8 @@
 #define WG 512
 #include "linalg_lds.h"
 
+typedef __attribute__((ext_vector_type(4))) float mfma_f32x4;
+
 struct NllLds {
   float* A;     // k * SA (SA = k+1 up-aligned to 4): lower K -> L/V ->
                 // K^-1 -> W0; upper: Kb cache
@@ -143,50 +145,62 @@ fused_expert_nll_kernel(const float* __restrict__ Xg,
   const float* Xe = Xg + (size_t)e * k * d;
   const float* ye = yg + (size_t)e * k;
 
-  // ---- A: stage X, y, s2 -------------------------------------------
-  for (int i = tid; i < k * d; i += WG) {
-    int a = i / d, j = i - a * d;
-    S.X[a * dp + j] = Xe[i];
-  }
-  for (int i = tid; i < k; i += WG) S.yb[i] = ye[i];
+  // ---- A: stage SCALED X' = X o s, y, s2 ----------------------------
+  // Scaled coordinates make the kernel build a plain GEMM
+  // (q = ||x'||^2 + ||a'||^2 - 2 x'.a') and the gradient contraction
+  // runs in scaled space, de-scaled by s2 at output (H phase).
   for (int j = tid; j < d; j += WG) {
     float s = scale[j];
     S.s2[j] = s * s;
   }
+  __syncthreads();
+  for (int i = tid; i < k * d; i += WG) {
+    int a = i / d, j = i - a * d;
+    S.X[a * dp + j] = Xe[i] * scale[j];
+  }
+  for (int i = tid; i < k; i += WG) S.yb[i] = ye[i];
   if (tid == 0) { *S.bad = 0; S.misc[0] = 0.0; }
   __syncthreads();
 
 PH(1);
-    // ---- B: lower = amp Kb + noise I; strict upper = Kb cache --------
+  // ---- B: lower = amp Kb + noise I; strict upper = Kb cache ---------
+  // MFMA form (round 2): q_ab = n_a + n_b - 2 x'_a . x'_b with the dot
+  // tiles on v_mfma_f32_16x16x4_f32 (exact fp32) — the elementwise
+  // version cost ~3 VALU issues per (pair, dim).  Row norms stage
+  // through S.alpha (free until phase E).
   {
-    const int nlow = k * (k + 1) / 2;
-    for (int f = tid; f < nlow; f += WG) {
-      int a, b;
-      tri_decode(f, a, b);
-      const float* xa = S.X + a * dp;
-      const float* xb = S.X + b * dp;
-      // float4 (ds_read_b128) over the 16-B-aligned X rows and s2
-      float4 q4 = {0.f, 0.f, 0.f, 0.f};
-      int j = 0;
-      for (; j + 3 < d; j += 4) {
-        const float4 va = *(const float4*)(xa + j);
-        const float4 vb = *(const float4*)(xb + j);
-        const float4 sv = *(const float4*)(S.s2 + j);
-        const float u0 = va.x - vb.x, u1 = va.y - vb.y;
-        const float u2 = va.z - vb.z, u3 = va.w - vb.w;
-        q4.x += sv.x * u0 * u0;
-        q4.y += sv.y * u1 * u1;
-        q4.z += sv.z * u2 * u2;
-        q4.w += sv.w * u3 * u3;
+    for (int a = tid; a < k; a += WG)
+      S.alpha[a] = dotv(S.X + (size_t)a * dp, S.X + (size_t)a * dp, 0, d);
+    __syncthreads();
+    const int nt = (k + 15) / 16;
+    const int ntri = nt * (nt + 1) / 2;
+    const int wave = tid >> 6;
+    const int l16 = lane & 15, kg = lane >> 4;
+    for (int t = wave; t < ntri; t += WG / 64) {
+      int ti, tj;
+      tri_decode(t, ti, tj);
+      mfma_f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+      const int ia = ti * 16 + l16;        // A-fragment row
+      const int jb = tj * 16 + l16;        // B-fragment column
+      for (int k0 = 0; k0 < d; k0 += 4) {
+        const int fk = k0 + kg;
+        const float av = (ia < k && fk < d) ? S.X[(size_t)ia * dp + fk]
+                                            : 0.f;
+        const float bv = (jb < k && fk < d) ? S.X[(size_t)jb * dp + fk]
+                                            : 0.f;
+        acc = __builtin_amdgcn_mfma_f32_16x16x4f32(av, bv, acc, 0, 0, 0);
       }
-      float q0 = (q4.x + q4.y) + (q4.z + q4.w);
-      for (; j < d; ++j) {
-        const float t = xa[j] - xb[j];
-        q0 += S.s2[j] * t * t;
+      // D map: row = (lane>>4)*4 + r, col = lane&15
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int gi = ti * 16 + kg * 4 + r;
+        const int gj = tj * 16 + l16;
+        if (gi >= k || gj >= k || gj > gi) continue;
+        float q = S.alpha[gi] + S.alpha[gj] - 2.0f * acc[r];
+        const float kb = __expf(-(q > 0.f ? q : 0.f));
+        S.A[(size_t)gi * SA + gj] = amp * kb + (gi == gj ? noise : 0.f);
+        if (gi != gj) S.A[(size_t)gj * SA + gi] = kb;   // Kb cache upper
       }
-      const float kb = __expf(-q0);
-      S.A[a * SA + b] = amp * kb + (a == b ? noise : 0.f);
-      if (a != b) S.A[b * SA + a] = kb;        // Kb cache in the upper
     }
   }
   __syncthreads();
@@ -310,7 +324,13 @@ PH(8);
                                   - (double)wx);
       }
       acc = wave_sum(acc);
-      if (lane == 0) out_contr[(size_t)e * d + d0 + j] = acc;
+      if (lane == 0) {
+        // contraction ran in scaled coordinates: contr' = s2_j * contr.
+        // s2_j == 0 (beta at its zero bound) => dK/dbeta_j == 0 and the
+        // host multiplies by beta_j anyway: emit 0 (the exact limit).
+        const double s2j = (double)S.s2[d0 + j];
+        out_contr[(size_t)e * d + d0 + j] = s2j > 0.0 ? acc / s2j : 0.0;
+      }
     }
     __syncthreads();
   }
