@@ -237,27 +237,54 @@ PH(3);
   const double yta = block_sum(part, S.red, tid);
 
 PH(5);
-  // ---- L: K^-1 = V^T V in place (lauum), ascending row blocks ------
-  // strictly j <= i so the Kb cache in the upper triangle survives.
-  // (A register-accumulated transposed-chunk variant measured SLOWER —
-  // 21.3 vs 18.5 us at k=100: the short per-chunk dots are overhead-
-  // bound and the unrolled accumulator array costs registers.  Negative
-  // result recorded in TODO.md; keep the strided dot4 form.)
-  for (int I = 0; I < nblk; ++I) {
-    const int ib = I * NB;
-    const int bs = min(NB, k - ib);
-    const int ncol = ib + bs;
-    for (int f = tid; f < bs * ncol; f += WG) {
-      const int r = f / ncol, j = f - r * ncol;
-      const int i = ib + r;
-      if (j > i) continue;
-      S.T[r * SA + j] = dot4(S.A + i, SA, S.A + j, SA, i, k);
+  // ---- L: K^-1 = V^T V (lauum) on f32 MFMA -------------------------
+  // SYRK-shaped: Kinv_ij = sum_c V[c][i] V[c][j] (V = S.A lower; the
+  // strict upper holds the Kb cache, so fragment reads are masked to
+  // c >= i).  Each wave holds its output tiles in registers across a
+  // barrier, then writes back — no staging buffer, no read/write race,
+  // Kb cache untouched.  (Round-2 history: a strided-dot4 version ran
+  // 18.3 us/expert at k=100; a register-accumulated chunk variant was
+  // SLOWER, 21.3 us; this MFMA form replaces both.)
+  {
+    const int nt = (k + 15) / 16;
+    const int ntri = nt * (nt + 1) / 2;
+    const int wave = tid >> 6;
+    const int l16 = lane & 15, kg = lane >> 4;
+    constexpr int MAXT = 5;                 // ceil(36 tiles / 8 waves)
+    mfma_f32x4 acc[MAXT];                   // compile-time indexed only
+#pragma unroll
+    for (int u = 0; u < MAXT; ++u) {
+      acc[u] = {0.f, 0.f, 0.f, 0.f};
+      const int t = wave + u * (WG / 64);
+      if (t >= ntri) continue;
+      int ti, tj;
+      tri_decode(t, ti, tj);
+      const int gi = ti * 16 + l16;         // A-fragment column of V
+      const int gj = tj * 16 + l16;         // B-fragment column of V
+      for (int c0 = 0; c0 < k; c0 += 4) {
+        const int cc = c0 + kg;
+        const float av = (cc < k && gi < k && cc >= gi)
+                             ? S.A[(size_t)cc * SA + gi] : 0.f;
+        const float bv = (cc < k && gj < k && cc >= gj)
+                             ? S.A[(size_t)cc * SA + gj] : 0.f;
+        acc[u] = __builtin_amdgcn_mfma_f32_16x16x4f32(av, bv, acc[u],
+                                                      0, 0, 0);
+      }
     }
-    __syncthreads();
-    for (int f = tid; f < bs * ncol; f += WG) {
-      const int r = f / ncol, j = f - r * ncol;
-      if (j > ib + r) continue;
-      S.A[(size_t)(ib + r) * SA + j] = S.T[r * SA + j];
+    __syncthreads();                        // all V reads complete
+#pragma unroll
+    for (int u = 0; u < MAXT; ++u) {
+      const int t = wave + u * (WG / 64);
+      if (t >= ntri) continue;
+      int ti, tj;
+      tri_decode(t, ti, tj);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int gi = ti * 16 + kg * 4 + r;
+        const int gj = tj * 16 + l16;
+        if (gi < k && gj < k && gj <= gi)
+          S.A[(size_t)gi * SA + gj] = acc[u][r];
+      }
     }
     __syncthreads();
   }
@@ -301,36 +328,60 @@ PH(7);
   __syncthreads();
 
 PH(8);
-  // ---- H: contraction, X^T-staged ----------------------------------
-  // contr_j = 2 sum_a x_aj^2 r_a - 2 sum_a x_aj (W0 X)_aj.  X columns
-  // are staged transposed through T (T[j][a] = X[a][d0+j], 16-B-aligned
-  // 36-float rows) so (W0 X)_aj = dotv(W0 row a, T row j) is a float4
-  // dot of two contiguous rows; wave w owns columns j = w (mod 8), lanes
-  // stride the row index — no barriers inside a chunk.
-  for (int d0 = 0; d0 < d; d0 += 32) {
-    const int dl = min(32, d - d0);
-    for (int f = tid; f < k * dl; f += WG) {
-      const int a = f / dl, j = f - a * dl;
-      S.T[j * SA + a] = S.X[a * dp + d0 + j];
+  // ---- H: contraction on f32 MFMA ----------------------------------
+  // contr'_j = 2 sum_a x'_aj (x'_aj r_a - (W0 X')_aj) with W0 = S.A
+  // (full symmetric after phase W) and X' the scaled coordinates: the
+  // W0 @ X' product is a [k, d] GEMM on 16x16x4 f32 tiles; each wave
+  // owns whole output tiles, folds its fragment against x' and r, and
+  // drops one fp32 partial per (row-tile, column) into T; a final pass
+  // sums the row-tile partials in fp64 and de-scales by s2_j.
+  {
+    const int nti = (k + 15) / 16;
+    const int ntj = (d + 15) / 16;
+    const int wave = tid >> 6;
+    const int l16 = lane & 63 & 15, kg = (lane & 63) >> 4;
+    for (int f = tid; f < nti * d; f += WG) S.T[f] = 0.f;
+    __syncthreads();
+    for (int t = wave; t < nti * ntj; t += WG / 64) {
+      const int ti = t / ntj, tj = t - ti * ntj;
+      mfma_f32x4 a = {0.f, 0.f, 0.f, 0.f};
+      const int fi = ti * 16 + l16;         // W0 row (A fragment)
+      const int fj = tj * 16 + l16;         // X' column (B fragment)
+      for (int c0 = 0; c0 < k; c0 += 4) {
+        const int cc = c0 + kg;
+        const float av = (fi < k && cc < k)
+                             ? S.A[(size_t)fi * SA + cc] : 0.f;
+        const float bv = (cc < k && fj < d)
+                             ? S.X[(size_t)cc * dp + fj] : 0.f;
+        a = __builtin_amdgcn_mfma_f32_16x16x4f32(av, bv, a, 0, 0, 0);
+      }
+      // fold: per fragment element (row gi, col gj):
+      //   part += 2 x'(gi,gj) * (x'(gi,gj) * r_gi - wx)
+      float part = 0.f;
+      const int gj = tj * 16 + l16;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int gi = ti * 16 + kg * 4 + r;
+        if (gi < k && gj < d) {
+          const float x = S.X[(size_t)gi * dp + gj];
+          part += 2.0f * x * (x * S.rrow[gi] - a[r]);
+        }
+      }
+      // reduce the 4 row-groups sharing this column: lanes l, l+16,
+      // l+32, l+48
+      part += __shfl_down(part, 32, 64);
+      part += __shfl_down(part, 16, 64);
+      if ((lane >> 4) == 0 && gj < d) S.T[ti * d + gj] = part;
     }
     __syncthreads();
-    const int ww = tid >> 6;
-    for (int j = ww; j < dl; j += 8) {
+    for (int j = tid; j < d; j += WG) {
       double acc = 0.0;
-      for (int a = lane; a < k; a += 64) {
-        const float x = S.T[j * SA + a];
-        const float wx = dotv(S.A + (size_t)a * SA, S.T + j * SA, 0, k);
-        acc += 2.0 * (double)x * ((double)x * (double)S.rrow[a]
-                                  - (double)wx);
-      }
-      acc = wave_sum(acc);
-      if (lane == 0) {
-        // contraction ran in scaled coordinates: contr' = s2_j * contr.
-        // s2_j == 0 (beta at its zero bound) => dK/dbeta_j == 0 and the
-        // host multiplies by beta_j anyway: emit 0 (the exact limit).
-        const double s2j = (double)S.s2[d0 + j];
-        out_contr[(size_t)e * d + d0 + j] = s2j > 0.0 ? acc / s2j : 0.0;
-      }
+      for (int ti = 0; ti < nti; ++ti) acc += (double)S.T[ti * d + j];
+      // contraction ran in scaled coordinates: contr' = s2_j * contr.
+      // s2_j == 0 (beta at its zero bound) => dK/dbeta_j == 0 and the
+      // host multiplies by beta_j anyway: emit 0 (the exact limit).
+      const double s2j = (double)S.s2[j];
+      out_contr[(size_t)e * d + j] = s2j > 0.0 ? acc / s2j : 0.0;
     }
     __syncthreads();
   }
