@@ -11,6 +11,8 @@
 #endif
 #define NB 32
 
+typedef __attribute__((ext_vector_type(4))) float mfma_f32x4_t;
+
 __device__ inline double wave_sum(double v) {
 #pragma unroll
   for (int off = 32; off; off >>= 1) v += __shfl_down(v, off, 64);
@@ -443,11 +445,35 @@ __device__ inline void chol_invert_lower(float* Abuf, float* Tbuf,
         Tbuf[r * 36 + c] = Abuf[(size_t)(t0 + r) * SA + jb + c];
       }
       __syncthreads();
-      // C2b: panel <- T * V_JJ^T : A[r][jb+c] = sum_{t<=c} T[r][t] V[c][t]
-      for (int f = tid; f < nr * bs; f += WG) {
-        int r = f / bs, c = f - r * bs;
-        Abuf[(size_t)(t0 + r) * SA + jb + c] =
-            dotv(Tbuf + r * 36, D + c * SA, 0, c + 1);
+      // C2b (MFMA): panel <- T @ V_JJ^T:
+      //   A[t0+r][jb+c] = sum_{t<=c} T[r][t] V_JJ[c][t]
+      // B-fragment = V_JJ[c][t] masked to its lower triangle; per-wave
+      // register tiles, written straight back to the panel.
+      {
+        const int w8 = tid >> 6;
+        const int fl16 = lane & 15, fkg = lane >> 4;
+        const int nti = (nr + 15) / 16, ntj = (bs + 15) / 16;
+        for (int t = w8; t < nti * ntj; t += WG / 64) {
+          const int ti = t / ntj, tj = t - ti * ntj;
+          mfma_f32x4_t a = {0.f, 0.f, 0.f, 0.f};
+          const int fr = ti * 16 + fl16;       // T row
+          const int fc = tj * 16 + fl16;       // V_JJ row (output col)
+          for (int c0 = 0; c0 < bs; c0 += 4) {
+            const int tt = c0 + fkg;
+            const float av = (fr < nr && tt < bs)
+                                 ? Tbuf[fr * 36 + tt] : 0.f;
+            const float bv = (fc < bs && tt <= fc)
+                                 ? D[(size_t)fc * SA + tt] : 0.f;
+            a = __builtin_amdgcn_mfma_f32_16x16x4f32(av, bv, a, 0, 0, 0);
+          }
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int gr = ti * 16 + fkg * 4 + r;
+            const int gc = tj * 16 + fl16;
+            if (gr < nr && gc < bs)
+              Abuf[(size_t)(t0 + gr) * SA + jb + gc] = a[r];
+          }
+        }
       }
       __syncthreads();
     }
@@ -458,25 +484,62 @@ __device__ inline void chol_invert_lower(float* Abuf, float* Tbuf,
   unsigned long long dt0 = 0;
   if (jclk && tid == 0) dt0 = wall_clock64();
   // ---- D: off-diagonal triangular inverse, in place, J right-to-left
-  // V_IJ = -(sum_{K=J+1..I} V_IK L_KJ) L_JJ^-1 ; rows fully parallel.
+  // V_IJ = -(sum_{K=J+1..I} V_IK L_KJ) L_JJ^-1, both halves as MFMA
+  // tiles: the U GEMM masks A-fragments to the lower triangle of the
+  // already-inverted trailing V (c <= row) and the second GEMM masks
+  // B-fragments to V_JJ's lower triangle (t >= j).
   for (int J = nblk - 2; J >= 0; --J) {
     const int jb = J * NB;
     const int bs = NB;                       // J < nblk-1 => full block
     const int t0 = jb + bs;
     const int nr = k - t0;
+    const int w8 = tid >> 6;
+    const int fl16 = lane & 15, fkg = lane >> 4;
+    const int nti = (nr + 15) / 16, ntj = (bs + 15) / 16;
     // U into T: U[r][t] = sum_{c=t0..t0+r} V[t0+r][c] * L[c][jb+t]
-    for (int f = tid; f < nr * bs; f += WG) {
-      int r = f / bs, t = f - r * bs;
-      const int row = t0 + r;
-      Tbuf[r * 36 + t] = dotm(Abuf + (size_t)row * SA,
-                              Abuf + jb + t, SA, t0, row + 1);
+    for (int t = w8; t < nti * ntj; t += WG / 64) {
+      const int ti = t / ntj, tj = t - ti * ntj;
+      mfma_f32x4_t a = {0.f, 0.f, 0.f, 0.f};
+      const int fr = t0 + ti * 16 + fl16;        // V row (A fragment)
+      const int fj = tj * 16 + fl16;             // L column (B fragment)
+      for (int c0 = t0; c0 < k; c0 += 4) {
+        const int cc = c0 + fkg;
+        const float av = (fr < k && cc < k && cc <= fr)
+                             ? Abuf[(size_t)fr * SA + cc] : 0.f;
+        const float bv = (cc < k && fj < bs)
+                             ? Abuf[(size_t)cc * SA + jb + fj] : 0.f;
+        a = __builtin_amdgcn_mfma_f32_16x16x4f32(av, bv, a, 0, 0, 0);
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int gr = ti * 16 + fkg * 4 + r;
+        const int gc = tj * 16 + fl16;
+        if (gr < nr && gc < bs) Tbuf[gr * 36 + gc] = a[r];
+      }
     }
     __syncthreads();
-    // V[row][jb+j] = - sum_{t>=j} U[r][t] * V_JJ[t][j]
-    for (int f = tid; f < nr * bs; f += WG) {
-      int r = f / bs, j = f - r * bs;
-      Abuf[(size_t)(t0 + r) * SA + jb + j] =
-          -dotm(Tbuf + r * 36, Abuf + (size_t)jb * SA + jb + j, SA, j, bs);
+    // V[t0+r][jb+j] = - sum_{t>=j} U[r][t] * V_JJ[t][j]
+    for (int t = w8; t < nti * ntj; t += WG / 64) {
+      const int ti = t / ntj, tj = t - ti * ntj;
+      mfma_f32x4_t a = {0.f, 0.f, 0.f, 0.f};
+      const int fr = ti * 16 + fl16;             // U row (A fragment)
+      const int fj = tj * 16 + fl16;             // V_JJ column (B frag)
+      for (int c0 = 0; c0 < bs; c0 += 4) {
+        const int tt = c0 + fkg;
+        const float av = (fr < nr && tt < bs)
+                             ? Tbuf[fr * 36 + tt] : 0.f;
+        const float bv = (tt < bs && fj < bs && tt >= fj)
+                             ? Abuf[(size_t)(jb + tt) * SA + jb + fj]
+                             : 0.f;
+        a = __builtin_amdgcn_mfma_f32_16x16x4f32(av, bv, a, 0, 0, 0);
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int gr = ti * 16 + fkg * 4 + r;
+        const int gc = tj * 16 + fl16;
+        if (gr < nr && gc < bs)
+          Abuf[(size_t)(t0 + gr) * SA + jb + gc] = -a[r];
+      }
     }
     __syncthreads();
   }
