@@ -353,8 +353,10 @@ std::vector<torch::Tensor> fused_laplace_newton(torch::Tensor X,
 
 bool fused_laplace_newton_supported(int64_t k, int64_t d) {
   if (k > 128 || d > k || k < 1) return false;
-  // EXACT mirror of lap_lds_bytes in laplace.hip
-  int64_t bytes = a16i(8 * 10) + 2 * a16i(4 * k * sa_of(k)) +
+  // EXACT mirror of lap_lds_bytes in laplace.hip (single-buffer design)
+  const int64_t dp4 = (d + 4) & ~(int64_t)3;
+  int64_t bytes = a16i(8 * 10) + a16i(4 * k * sa_of(k)) +
+                  a16i(4 * k * dp4) +
                   a16i(4 * tsz_of(k)) + 8 * a16i(4 * k) + a16i(4 * d) + 16;
   return bytes <= 160 * 1024;
 }
@@ -403,7 +405,9 @@ std::vector<torch::Tensor> fused_laplace_evidence(torch::Tensor X,
 bool fused_laplace_evidence_supported(int64_t k, int64_t d) {
   if (k > 128 || d > k || k < 1) return false;
   // EXACT mirror of lap_lds_bytes(k, d, ev=1) in laplace.hip
-  int64_t bytes = a16i(8 * 10) + 2 * a16i(4 * k * sa_of(k)) +
+  const int64_t dp4 = (d + 4) & ~(int64_t)3;
+  int64_t bytes = a16i(8 * 10) + a16i(4 * k * sa_of(k)) +
+                  a16i(4 * k * dp4) +
                   a16i(4 * tsz_of(k)) + 8 * a16i(4 * k) + a16i(4 * d) + 16 +
                   2 * a16i(4 * k) + a16i(4 * d) + a16i(8 * (d + 2));
   return bytes <= 160 * 1024;

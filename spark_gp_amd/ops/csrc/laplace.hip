@@ -33,8 +33,10 @@
 #include "linalg_lds.h"
 
 struct LapLds {
-  float* KB;    // k * SA: full K = amp*Kb + noise*I (both triangles)
-  float* A;     // k * SA: X staging, then B -> L -> V per iteration
+  float* X;     // k * dp4: raw features (16-B-aligned rows)
+  float* A;     // k * SA: strict upper = Kb cache (kb, no amp/noise;
+                // kb_aa = 1 implicit); lower = B -> L -> V per iteration
+                // and Binv -> W0 in the evidence tail
   float* T;     // max(k*36, 32*SA, 448) scratch for the chol machinery
   float* yb;    // k
   float* fb;    // k   latent
@@ -72,10 +74,15 @@ static __host__ __device__ inline size_t lap_a16(size_t n) {
   return (n + 15) & ~(size_t)15;
 }
 
+static __host__ __device__ inline size_t lap_dp4(int d) {
+  return (size_t)((d + 4) & ~3);
+}
+
 static __host__ __device__ inline size_t lap_lds_bytes(int k, int d,
                                                        int ev = 0) {
   size_t off = lap_a16(sizeof(double) * 10);
-  off += 2 * lap_a16(sizeof(float) * (size_t)k * lap_sa(k));  // KB + A
+  off += lap_a16(sizeof(float) * (size_t)k * lap_sa(k));      // A
+  off += lap_a16(sizeof(float) * (size_t)k * lap_dp4(d));     // X
   off += lap_a16(sizeof(float) * lap_tsz(k));
   off += 8 * lap_a16(sizeof(float) * k);
   off += lap_a16(sizeof(float) * d);
@@ -94,8 +101,8 @@ __device__ inline LapLds lap_carve(char* base, int k, int d) {
   char* p = base;
   L.red = (double*)p;  L.misc = L.red + 8;
   p += lap_a16(sizeof(double) * 10);
-  L.KB = (float*)p;    p += lap_a16(sizeof(float) * (size_t)k * lap_sa(k));
   L.A = (float*)p;     p += lap_a16(sizeof(float) * (size_t)k * lap_sa(k));
+  L.X = (float*)p;     p += lap_a16(sizeof(float) * (size_t)k * lap_dp4(d));
   L.T = (float*)p;     p += lap_a16(sizeof(float) * lap_tsz(k));
   L.yb = (float*)p;    p += lap_a16(sizeof(float) * k);
   L.fb = (float*)p;    p += lap_a16(sizeof(float) * k);
@@ -114,6 +121,26 @@ __device__ inline LapLds lap_carve(char* base, int k, int d) {
   }
   L.bad = (int*)p;
   return L;
+}
+
+// K = amp * kb + noise * I with kb cached in A's STRICT UPPER triangle
+// (kb_aa = 1 implicit): row-i dot against a contiguous vector.  The b < i
+// half reads the upper entries [b][i] (stride SA); the b > i half is the
+// contiguous row tail.  Never touches A's lower (B/L/V live there).
+__device__ inline float lap_kdot(const float* A, int SA, int k, int i,
+                                 const float* vec, float amp, float noise) {
+  const float s = dot4(A + i, SA, vec, 1, 0, i)
+                + dotv(A + (size_t)i * SA, vec, i + 1, k);
+  return amp * (s + vec[i]) + noise * vec[i];
+}
+
+// same against a strided column (the 8-wide chunk buffers); returns the
+// NOISELESS product Kc q = amp * (kb q)
+__device__ inline float lap_kcdot(const float* A, int SA, int k, int a,
+                                  const float* q, int sq, float amp) {
+  const float s = dot4(A + a, SA, q, sq, 0, a)
+                + dotm(A + (size_t)a * SA, q, sq, a + 1, k);
+  return amp * (s + q[a * sq]);
 }
 
 __device__ inline double log_sigmoid(double v) {
@@ -144,11 +171,12 @@ fused_laplace_kernel(const float* __restrict__ Xg,   // [E, k, d]
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const float* Xe = Xg + (size_t)e * k * d;
+  const int dp = (int)lap_dp4(d);
 
-  // ---- stage X (into A), y, f, s2 ----------------------------------
+  // ---- stage X, y, f, s2 -------------------------------------------
   for (int i = tid; i < k * d; i += WG) {
     int a = i / d, j = i - a * d;
-    S.A[a * SA + j] = Xe[i];
+    S.X[a * dp + j] = Xe[i];
   }
   for (int i = tid; i < k; i += WG) {
     S.yb[i] = yg[(size_t)e * k + i];
@@ -162,26 +190,37 @@ fused_laplace_kernel(const float* __restrict__ Xg,   // [E, k, d]
   if (tid == 0) *S.bad = 0;
   __syncthreads();
 
-  // ---- KB = amp * exp(-q) + noise I (full symmetric) ----------------
+  // ---- Kb cache into A's STRICT UPPER (kb = exp(-q); kb_aa = 1) ----
+  // Single-buffer design (round 2): the old separate full-K buffer cost
+  // a second k x SA region and held the whole kernel at 1 WG/CU; with
+  // the expert-NLL upper-cache idiom the kernel fits 2 WGs/CU.
   {
     const int nlow = k * (k + 1) / 2;
     for (int f = tid; f < nlow; f += WG) {
       int a, b;
       tri_decode(f, a, b);
-      const float* xa = S.A + a * SA;
-      const float* xb = S.A + b * SA;
-      float q0 = 0.f, q1 = 0.f;
+      if (a == b) continue;
+      const float* xa = S.X + (size_t)a * dp;
+      const float* xb = S.X + (size_t)b * dp;
+      float4 q4 = {0.f, 0.f, 0.f, 0.f};
       int j = 0;
-      for (; j + 1 < d; j += 2) {
-        float u0 = xa[j] - xb[j];
-        float u1 = xa[j + 1] - xb[j + 1];
-        q0 += S.s2[j] * u0 * u0;
-        q1 += S.s2[j + 1] * u1 * u1;
+      for (; j + 3 < d; j += 4) {
+        const float4 va = *(const float4*)(xa + j);
+        const float4 vb = *(const float4*)(xb + j);
+        const float4 sv = *(const float4*)(S.s2 + j);
+        const float u0 = va.x - vb.x, u1 = va.y - vb.y;
+        const float u2 = va.z - vb.z, u3 = va.w - vb.w;
+        q4.x += sv.x * u0 * u0;
+        q4.y += sv.y * u1 * u1;
+        q4.z += sv.z * u2 * u2;
+        q4.w += sv.w * u3 * u3;
       }
-      if (j < d) { float u = xa[j] - xb[j]; q0 += S.s2[j] * u * u; }
-      const float kv = amp * __expf(-(q0 + q1)) + (a == b ? noise : 0.f);
-      S.KB[a * SA + b] = kv;
-      if (a != b) S.KB[b * SA + a] = kv;
+      float q0 = (q4.x + q4.y) + (q4.z + q4.w);
+      for (; j < d; ++j) {
+        const float u = xa[j] - xb[j];
+        q0 += S.s2[j] * u * u;
+      }
+      S.A[(size_t)b * SA + a] = __expf(-q0);      // upper: b < a
     }
   }
   __syncthreads();
@@ -205,17 +244,19 @@ fused_laplace_kernel(const float* __restrict__ Xg,   // [E, k, d]
     __syncthreads();
     // t1 = K b
     for (int i = tid; i < k; i += WG)
-      S.t1[i] = dotv(S.KB + (size_t)i * SA, S.bv, 0, k);
+      S.t1[i] = lap_kdot(S.A, SA, k, i, S.bv, amp, noise);
     __syncthreads();
-    // u = sqw * (K b) into t1; build B (lower) into A
+    // u = sqw * (K b) into t1; build B (lower) into A from the upper cache
     for (int i = tid; i < k; i += WG) S.t1[i] *= S.sqw[i];
     {
       const int nlow = k * (k + 1) / 2;
       for (int f = tid; f < nlow; f += WG) {
         int a, b;
         tri_decode(f, a, b);
+        const float kv = (a == b) ? (amp + noise)
+                                  : amp * S.A[(size_t)b * SA + a];
         S.A[(size_t)a * SA + b] =
-            (a == b ? 1.f : 0.f) + S.sqw[a] * S.KB[(size_t)a * SA + b] * S.sqw[b];
+            (a == b ? 1.f : 0.f) + S.sqw[a] * kv * S.sqw[b];
       }
     }
     __syncthreads();
@@ -234,7 +275,7 @@ fused_laplace_kernel(const float* __restrict__ Xg,   // [E, k, d]
     __syncthreads();
     // t1 = K a;  f_cand (t2) = (1-s) f + s K a
     for (int i = tid; i < k; i += WG)
-      S.t1[i] = dotv(S.KB + (size_t)i * SA, S.av, 0, k);
+      S.t1[i] = lap_kdot(S.A, SA, k, i, S.av, amp, noise);
     __syncthreads();
     const float sf = (float)step;
     double part = 0.0;
@@ -282,8 +323,10 @@ fused_laplace_kernel(const float* __restrict__ Xg,   // [E, k, d]
       for (int f = tid; f < nlow; f += WG) {
         int a, b;
         tri_decode(f, a, b);
-        S.A[(size_t)a * SA + b] = (a == b ? 1.f : 0.f) +
-            S.sqw[a] * S.KB[(size_t)a * SA + b] * S.sqw[b];
+        const float kv = (a == b) ? (amp + noise)
+                                  : amp * S.A[(size_t)b * SA + a];
+        S.A[(size_t)a * SA + b] =
+            (a == b ? 1.f : 0.f) + S.sqw[a] * kv * S.sqw[b];
       }
     }
     __syncthreads();
@@ -292,7 +335,7 @@ fused_laplace_kernel(const float* __restrict__ Xg,   // [E, k, d]
   if (EV && !*S.bad) {
     // a = b - sqw V^T V (sqw (K b));  fc = K a;  psi;  logZ
     for (int i = tid; i < k; i += WG)
-      S.t1[i] = dotv(S.KB + (size_t)i * SA, S.bv, 0, k) * S.sqw[i];
+      S.t1[i] = lap_kdot(S.A, SA, k, i, S.bv, amp, noise) * S.sqw[i];
     __syncthreads();
     for (int i = tid; i < k; i += WG)
       S.t2[i] = dotv(S.A + (size_t)i * SA, S.t1, 0, i + 1);
@@ -305,7 +348,7 @@ fused_laplace_kernel(const float* __restrict__ Xg,   // [E, k, d]
     __syncthreads();
     double part = 0.0;
     for (int i = tid; i < k; i += WG) {
-      const float fc = dotv(S.KB + (size_t)i * SA, S.av, 0, k);
+      const float fc = lap_kdot(S.A, SA, k, i, S.av, amp, noise);
       part += -0.5 * (double)S.av[i] * (double)fc
               + log_sigmoid((double)((2.f * S.yb[i] - 1.f) * fc));
     }
@@ -320,28 +363,34 @@ fused_laplace_kernel(const float* __restrict__ Xg,   // [E, k, d]
     }
     for (int j = tid; j < d + 2; j += WG) S.gout[j] = 0.0;
     __syncthreads();
-    // diagKRK_i = || V (sqw o K_:,i) ||^2  (V lower in A; K symmetric)
-    for (int f = tid; f < k * k; f += WG) {
-      const int i = f / k, j = f - i * k;
-      const float* vr = S.A + (size_t)j * SA;
-      const float* kr = S.KB + (size_t)i * SA;
-      float z0 = 0.f, z1 = 0.f;
-      int c = 0;
-      for (; c + 1 <= j; c += 2) {
-        z0 += vr[c] * S.sqw[c] * kr[c];
-        z1 += vr[c + 1] * S.sqw[c + 1] * kr[c + 1];
+    // diagKRK_i = || V (sqw o K_:,i) ||^2  (V lower in A; K symmetric).
+    // K rows come from the upper cache through a staged 8-row block in T
+    // so the inner dots stay contiguous-contiguous.
+    for (int i0 = 0; i0 < k; i0 += 8) {
+      const int ib = min(8, k - i0);
+      for (int f = tid; f < ib * k; f += WG) {
+        const int il = f / k, c = f - il * k;
+        const int i = i0 + il;
+        const float kv = (c == i) ? (amp + noise)
+                         : amp * ((c < i) ? S.A[(size_t)c * SA + i]
+                                          : S.A[(size_t)i * SA + c]);
+        S.T[il * SA + c] = S.sqw[c] * kv;
       }
-      if (c <= j) z0 += vr[c] * S.sqw[c] * kr[c];
-      const float z = z0 + z1;
-      atomicAdd(&S.t2[i], z * z);
+      __syncthreads();
+      for (int f = tid; f < ib * k; f += WG) {
+        const int il = f / k, j = f - il * k;
+        const float z = dotv(S.A + (size_t)j * SA, S.T + il * SA, 0, j + 1);
+        atomicAdd(&S.t2[i0 + il], z * z);
+      }
+      __syncthreads();
     }
-    __syncthreads();
-    // s2 vector; then u0 = K v - nu v into t2
+    // s2 vector (K_ii = amp + noise); then u0 = K v - nu v into t2
     for (int i = tid; i < k; i += WG)
-      S.s2v[i] = -0.5f * (S.KB[(size_t)i * SA + i] - S.t2[i]) * S.pi[i];
+      S.s2v[i] = -0.5f * (amp + noise - S.t2[i]) * S.pi[i];
     __syncthreads();
     for (int i = tid; i < k; i += WG)
-      S.t1[i] = dotv(S.KB + (size_t)i * SA, S.vv, 0, k) - noise * S.vv[i];
+      S.t1[i] = lap_kdot(S.A, SA, k, i, S.vv, amp, noise)
+                - noise * S.vv[i];
     __syncthreads();
     for (int i = tid; i < k; i += WG) S.t2[i] = S.t1[i];   // u0
     __syncthreads();
@@ -358,7 +407,7 @@ fused_laplace_kernel(const float* __restrict__ Xg,   // [E, k, d]
         const int j = j0 + cc;
         float x = 0.f, w1 = 0.f, w2 = 0.f;
         if (cc < cl && j < d) {
-          x = Xe[(size_t)a * d + j];
+          x = S.X[(size_t)a * dp + j];
           w1 = x * S.vv[a];
           w2 = x * x * S.vv[a];
         }
@@ -370,14 +419,12 @@ fused_laplace_kernel(const float* __restrict__ Xg,   // [E, k, d]
       // U1 = Kc (X_j o v) into bmb; then U2 = Kc (X_j^2 o v) into wk1
       for (int f = tid; f < k * 8; f += WG) {
         const int a = f >> 3, cc = f & 7;
-        bmb[f] = dotm(S.KB + (size_t)a * SA, wk1 + cc, 8, 0, k)
-                 - noise * wk1[f];
+        bmb[f] = lap_kcdot(S.A, SA, k, a, wk1 + cc, 8, amp);
       }
       __syncthreads();
       for (int f = tid; f < k * 8; f += WG) {
         const int a = f >> 3, cc = f & 7;
-        const float u2 = dotm(S.KB + (size_t)a * SA, wk2 + cc, 8, 0, k)
-                         - noise * wk2[f];
+        const float u2 = lap_kcdot(S.A, SA, k, a, wk2 + cc, 8, amp);
         const int j = j0 + cc;
         float bm;
         if (j < d) {
@@ -419,7 +466,8 @@ fused_laplace_kernel(const float* __restrict__ Xg,   // [E, k, d]
       __syncthreads();
       for (int f = tid; f < k * 8; f += WG) {
         const int a = f >> 3, cc = f & 7;
-        wk2[f] = bmb[f] - dotm(S.KB + (size_t)a * SA, qb + cc, 8, 0, k);
+        wk2[f] = bmb[f] - (lap_kcdot(S.A, SA, k, a, qb + cc, 8, amp)
+                           + noise * qb[a * 8 + cc]);
       }
       __syncthreads();
       // gout[j] += sum_a s2v[a] * s3[a][cc]   (wave cc owns column cc)
@@ -457,14 +505,9 @@ fused_laplace_kernel(const float* __restrict__ Xg,   // [E, k, d]
         }
         __syncthreads();
       }
-      const int nlow = k * (k + 1) / 2;
-      for (int f = tid; f < nlow; f += WG) {
-        int a, b;
-        tri_decode(f, a, b);
-        if (a != b) S.A[(size_t)b * SA + a] = S.A[(size_t)a * SA + b];
-      }
-      __syncthreads();
     }
+    // (no mirror: the W phase reads Binv from the lower triangle only,
+    // and the strict upper must keep the Kb cache until W consumes it)
 
     // ---- W phase: scalars + W0 = (a a^T - R) o Kb in place -----------
     double trR_p = 0.0, sRKb_p = 0.0, aKba_p = 0.0, a2_p = 0.0;
@@ -475,8 +518,7 @@ fused_laplace_kernel(const float* __restrict__ Xg,   // [E, k, d]
         tri_decode(f, a, b);
         const float binv = S.A[(size_t)a * SA + b];
         const float R = S.sqw[a] * S.sqw[b] * binv;
-        const float kb = (S.KB[(size_t)a * SA + b] - (a == b ? noise : 0.f))
-                         / amp;
+        const float kb = (a == b) ? 1.f : S.A[(size_t)b * SA + a];
         const float g = S.av[a] * S.av[b] - R;
         const float w = g * kb;
         if (a == b) {
@@ -517,7 +559,7 @@ fused_laplace_kernel(const float* __restrict__ Xg,   // [E, k, d]
       const int cl = min(8, d - d0);
       for (int f = tid; f < k * 8; f += WG) {
         const int a = f >> 3, cc = f & 7;
-        qb[f] = (cc < cl) ? Xe[(size_t)a * d + d0 + cc] : 0.f;
+        qb[f] = (cc < cl) ? S.X[(size_t)a * dp + d0 + cc] : 0.f;
       }
       __syncthreads();
       for (int f = tid; f < k * 8; f += WG) {
