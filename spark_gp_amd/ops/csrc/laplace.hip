@@ -150,7 +150,7 @@ __device__ inline double log_sigmoid(double v) {
 }
 
 template <bool EV>
-__global__ void __launch_bounds__(WG)
+__global__ void __launch_bounds__(WG, 4)   // 4 waves/SIMD = 2 WGs/CU
 fused_laplace_kernel(const float* __restrict__ Xg,   // [E, k, d]
                      const float* __restrict__ yg,   // [E, k]
                      float* __restrict__ fg,         // [E, k] inout
