@@ -7,7 +7,14 @@
 //    both operand tiles stored k-major so every fragment load is one
 //    ds_read_b128; split-K over the chunk for occupancy; fp32 atomic
 //    accumulation into the output.
-//  * colsum_gemv        — Ky[m] += K_nm^T y (fp64 accumulation).
+//  * cross_mfma         — the round-2 PPA fast path: sqdist via
+//    ||x'||^2 + ||a'||^2 - 2 x'.a' on f32 matrix cores, transposed hi/lo
+//    outputs only, Ky += K^T y fused into the epilogue.
+//  * syrk_bf16_sync     — k-synchronized persistent SYRK for large m
+//    (one block owns one tile; bounded best-effort pacing).
+//  * colsum_gemv        — Ky[m] += K_nm^T y (fp64 accumulation;
+//    superseded on the PPA path by the fused epilogues, kept as the
+//    standalone op).
 //
 // Replaces ProjectedGaussianProcessHelper.scala:20-36's per-expert
 // crossKernel + breeze gemm accumulation.

@@ -9,7 +9,10 @@
 // between stages, and WITHOUT materializing the [d, k, k] derivative tensor:
 //
 //   B  lower(A) = amp*Kb + noise*I,  strict upper(A) = Kb (cached base
-//      kernel — read back in phase W, never recomputed)
+//      kernel — read back in phase W, never recomputed); round 2: the
+//      sqdist is q = n_a + n_b - 2 x'.a' over pre-SCALED coordinates on
+//      f32 MFMA tiles (the gradient contraction runs in scaled space and
+//      is de-scaled by s2 at output)
 //   C  blocked right-looking Cholesky, IN PLACE, with look-ahead: per
 //      8-column sub-block q of each 32-column block, wave 0 factors AND
 //      inverts the 8x8 sub-diagonal (row-per-lane on 8 lanes, cross-lane
@@ -25,12 +28,13 @@
 //      (right-to-left column blocks, rows fully parallel:
 //       V_IJ = -(sum_{K>J} V_IK L_KJ) L_JJ^-1).
 //   E  alpha = V^T (V y);  nll = 1/2 y.alpha + 1/2 logdet
-//   L  in-place lauum: K^-1 = V^T V (ascending row blocks through a temp
-//      row buffer; strictly lower+diagonal so the Kb cache survives)
+//   L  lauum K^-1 = V^T V on f32 MFMA (masked fragments, register tiles
+//      across a barrier; strictly lower+diagonal so the Kb cache survives)
 //   W  W0 = (alpha alpha^T - K^-1) o Kb (Kb from the upper-triangle cache);
 //      trG / sum(W0) accumulated on the fly
-//   H  gradient contraction per input dim (the K5 fusion):
-//      contr_j = 2 sum_a x_aj^2 r_a - 2 sum_a x_aj (W0 X)_aj
+//   H  gradient contraction per input dim (the K5 fusion)
+//      on f32 MFMA: W0 @ X' as 16x16x4 tiles, folded per fragment into
+//      per-column partials; contr de-scaled by s2 at output
 //
 // Inner products with a contiguous operand are float4 (ds_read_b128)
 // vectorized over 16-B-aligned LDS rows (strides padded to multiples of 4
@@ -57,7 +61,7 @@ struct NllLds {
   float* A;     // k * SA (SA = k+1 up-aligned to 4): lower K -> L/V ->
                 // K^-1 -> W0; upper: Kb cache
   float* T;     // temp: max(k*36, 32*SA, 448)
-  float* X;     // k * (d+1) raw features
+  float* X;     // k * dp4 SCALED features x' = x o s (16-B-aligned rows)
   float* yb;    // k
   float* alpha; // k
   float* tvec;  // k

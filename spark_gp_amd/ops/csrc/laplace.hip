@@ -17,14 +17,22 @@
 //
 // Outputs: updated latent f (in place), psi, sum log diag L, per-expert
 // Newton iteration count, bad flag (fp32 breakdown -> host falls back to
-// the batched torch path for that expert).  The evidence/gradient pass
-// (Algorithm 5.1) runs on the torch path AT the converged f in contraction
-// form (torch_backend.laplace_evidence_compiled — one batched fp64
-// Cholesky, no [E, p, k, k] tensor, reference-exact semantics); if any
-// expert went bad, the whole batch falls back to the torch Newton loop
-// warm-started from f.
+// the batched torch path for that expert).  In EV mode the Algorithm 5.1
+// evidence + FULL gradient (K11) run in the same launch at the converged
+// latent (see the evidence tail below); outside EV mode — or on any bad
+// expert, or for tolerances under LAPLACE_MIN_TOL — the torch
+// contraction-form path finishes with reference-exact fp64 semantics.
 //
-// Constraints: k <= 128, d <= k (X stages through the A buffer).
+// Single-buffer design (round 2): ONE k x SA working matrix — the base
+// kernel kb lives in its STRICT UPPER triangle (kb_aa = 1 implicit) and
+// survives every phase that works on the lower (B -> L -> V -> Binv);
+// K = amp*kb + noise*I is reconstructed on the fly (lap_kdot/lap_kcdot
+// split each row dot into a strided upper half and a contiguous tail).
+// With __launch_bounds__(512, 4) the kernel runs 2 workgroups per CU —
+// the old separate-K version was 1/CU twice over (2 x k x SA LDS AND
+// 232 VGPRs) and ran the 1M x 16 GPC fit ~1.6x slower.
+//
+// Constraints: k <= 128, d <= k; LDS budget checked host-side.
 
 #include <hip/hip_runtime.h>
 #include <math.h>
